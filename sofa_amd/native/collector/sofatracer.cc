@@ -1,0 +1,520 @@
+// libsofatracer — rocprofiler-sdk tool library: the MI355X-native replacement
+// for the reference's system-wide nvprof/CUPTI capture
+// (cyliustack/sofa bin/sofa_record.py:217-242 + nvvp/sqlite parsing in
+// bin/sofa_preprocess.py:1340-1543).
+//
+// Loaded into every GPU process of the profiled command via
+// ROCP_TOOL_LIBRARIES (rocprofiler-register picks it up when the HIP/HSA
+// runtime initializes).  Captures, with buffered (lossless) tracing:
+//   * kernel dispatches (agent, queue, kernel_id, grid/workgroup, LDS/scratch)
+//   * async memory copies (H2D/D2H/D2D/P2P with src/dst device)
+//   * HIP runtime API spans (optional, SOFA_TRACE_HIP_API=1)
+//   * memory allocations (optional)
+// and with callback tracing:
+//   * RCCL API calls with full args (count, dtype, comm, stream, peer/root) —
+//     the basis for per-xGMI-link collective attribution in sofa_analyze
+//   * code-object kernel-symbol registration (kernel_id -> mangled name)
+//   * roctx markers (MARKER_CORE API) for user annotations.
+//
+// Output: one binary SGT file per process in $SOFA_LOGDIR (sgt_format.h);
+// parsed vectorized (numpy structured dtypes) by sofa_amd.preprocess.gpu.
+// Timestamps are rocprofiler's ns clock; REC_CLOCK records at init+fini give
+// the (REALTIME, MONOTONIC_RAW, rocp) correlation used by preprocess to place
+// GPU events on the unified timeline (SURVEY.md §7 "three-clock sync").
+//
+// Env knobs:
+//   SOFA_LOGDIR        output directory (default: cwd)
+//   SOFA_TRACE_HIP_API 1/0 (default 1)
+//   SOFA_TRACE_RCCL    1/0 (default 1)
+//   SOFA_TRACE_ALLOC   1/0 (default 0)
+//   SOFA_GPU_BUFFER_MB per-process SDK buffer MiB (default 64)
+
+#include <rocprofiler-sdk/registration.h>
+#include <rocprofiler-sdk/rocprofiler.h>
+#include <rocprofiler-sdk/rccl.h>
+#include <rocprofiler-sdk/marker/api_id.h>
+
+#include <unistd.h>
+
+#include <atomic>
+#include <cstdint>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <ctime>
+#include <mutex>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+#include "sgt_format.h"
+
+namespace {
+
+rocprofiler_client_id_t* g_client_id = nullptr;
+rocprofiler_context_id_t g_ctx = {0};
+rocprofiler_buffer_id_t g_buffer = {};
+
+FILE* g_out = nullptr;
+std::mutex g_mutex;
+std::atomic<uint64_t> g_n_records{0};
+
+// agent handle -> logical GPU index (or -1 for CPU agents)
+std::unordered_map<uint64_t, int32_t> g_agent_device;
+
+bool env_flag(const char* name, bool dflt) {
+  const char* v = getenv(name);
+  if (!v || !*v) return dflt;
+  return !(v[0] == '0' || v[0] == 'n' || v[0] == 'N' || v[0] == 'f' || v[0] == 'F');
+}
+
+uint64_t host_ns(clockid_t c) {
+  struct timespec ts;
+  clock_gettime(c, &ts);
+  return uint64_t(ts.tv_sec) * 1000000000ull + ts.tv_nsec;
+}
+
+void write_raw(const void* p, size_t n) {
+  std::lock_guard<std::mutex> lk(g_mutex);
+  if (g_out) fwrite(p, 1, n, g_out);
+}
+
+void write_name_rec(uint16_t type, uint64_t id, const char* name) {
+  size_t len = name ? strlen(name) : 0;
+  size_t total = (sizeof(sgt::NameRec) + len + 1 + 7) & ~size_t(7);
+  std::vector<char> buf(total, 0);
+  auto* rec = reinterpret_cast<sgt::NameRec*>(buf.data());
+  rec->h = {type, static_cast<uint16_t>(total), 0};
+  rec->id = id;
+  if (len) memcpy(buf.data() + sizeof(sgt::NameRec), name, len);
+  write_raw(buf.data(), total);
+}
+
+void write_opname_rec(uint32_t kind, uint32_t op, const char* name) {
+  size_t len = name ? strlen(name) : 0;
+  size_t total = (sizeof(sgt::OpNameRec) + len + 1 + 7) & ~size_t(7);
+  std::vector<char> buf(total, 0);
+  auto* rec = reinterpret_cast<sgt::OpNameRec*>(buf.data());
+  rec->h = {sgt::REC_OPNAME, static_cast<uint16_t>(total), 0};
+  rec->kind = kind;
+  rec->op = op;
+  if (len) memcpy(buf.data() + sizeof(sgt::OpNameRec), name, len);
+  write_raw(buf.data(), total);
+}
+
+void write_clock_rec() {
+  sgt::ClockRec rec{};
+  rec.h = {sgt::REC_CLOCK, sizeof(sgt::ClockRec), 0};
+  rocprofiler_timestamp_t ts = 0;
+  rocprofiler_get_timestamp(&ts);
+  rec.realtime_ns = host_ns(CLOCK_REALTIME);
+  rec.monotonic_raw_ns = host_ns(CLOCK_MONOTONIC_RAW);
+  rec.rocp_ns = ts;
+  write_raw(&rec, sizeof(rec));
+}
+
+int32_t agent_device(rocprofiler_agent_id_t id) {
+  auto it = g_agent_device.find(id.handle);
+  return it == g_agent_device.end() ? -1 : it->second;
+}
+
+// RCCL datatype element sizes (ncclDataType_t order: int8, uint8, int32,
+// uint32, int64, uint64, half, float, double, bf16, fp8e4m3, fp8e5m2)
+uint32_t nccl_elem_size(uint32_t dt) {
+  static const uint32_t sz[] = {1, 1, 4, 4, 8, 8, 2, 4, 8, 2, 1, 1};
+  return dt < sizeof(sz) / sizeof(sz[0]) ? sz[dt] : 0;
+}
+
+// ---------------------------------------------------------------- callbacks
+
+void code_object_callback(rocprofiler_callback_tracing_record_t record,
+                          rocprofiler_user_data_t*, void*) {
+  if (record.kind == ROCPROFILER_CALLBACK_TRACING_CODE_OBJECT &&
+      record.operation == ROCPROFILER_CODE_OBJECT_DEVICE_KERNEL_SYMBOL_REGISTER &&
+      record.phase == ROCPROFILER_CALLBACK_PHASE_LOAD) {
+    auto* data = static_cast<
+        rocprofiler_callback_tracing_code_object_kernel_symbol_register_data_t*>(
+        record.payload);
+    write_name_rec(sgt::REC_KERNEL_NAME, data->kernel_id, data->kernel_name);
+  } else if (record.kind == ROCPROFILER_CALLBACK_TRACING_CODE_OBJECT &&
+             record.operation == ROCPROFILER_CODE_OBJECT_LOAD &&
+             record.phase == ROCPROFILER_CALLBACK_PHASE_UNLOAD) {
+    auto status = rocprofiler_flush_buffer(g_buffer);
+    (void) status;
+  }
+}
+
+// RCCL callback tracing: start/stop phases; we record the span + args.
+struct RcclPending {
+  uint64_t start_ns;
+};
+thread_local std::unordered_map<uint64_t, RcclPending> t_rccl_pending;
+
+void rccl_callback(rocprofiler_callback_tracing_record_t record,
+                   rocprofiler_user_data_t* user_data, void*) {
+  if (record.kind != ROCPROFILER_CALLBACK_TRACING_RCCL_API) return;
+  rocprofiler_timestamp_t now = 0;
+  rocprofiler_get_timestamp(&now);
+  if (record.phase == ROCPROFILER_CALLBACK_PHASE_ENTER) {
+    user_data->value = now;
+    return;
+  }
+  if (record.phase != ROCPROFILER_CALLBACK_PHASE_EXIT) return;
+
+  auto* data =
+      static_cast<rocprofiler_callback_tracing_rccl_api_data_t*>(record.payload);
+  sgt::RcclRec rec{};
+  rec.h = {sgt::REC_RCCL, sizeof(sgt::RcclRec), 0};
+  rec.start_ns = user_data->value;
+  rec.end_ns = now;
+  rec.corr_id = record.correlation_id.internal;
+  rec.tid = static_cast<uint32_t>(record.thread_id);
+  rec.op = record.operation;
+  rec.peer_or_root = -1;
+  rec.device = 0;
+
+  const auto& a = data->args;
+  switch (record.operation) {
+    case ROCPROFILER_RCCL_API_ID_ncclAllReduce:
+      rec.count = a.ncclAllReduce.count;
+      rec.datatype = a.ncclAllReduce.datatype;
+      rec.comm = reinterpret_cast<uint64_t>(a.ncclAllReduce.comm);
+      rec.stream = reinterpret_cast<uint64_t>(a.ncclAllReduce.stream);
+      break;
+    case ROCPROFILER_RCCL_API_ID_ncclAllGather:
+      rec.count = a.ncclAllGather.sendcount;
+      rec.datatype = a.ncclAllGather.datatype;
+      rec.comm = reinterpret_cast<uint64_t>(a.ncclAllGather.comm);
+      rec.stream = reinterpret_cast<uint64_t>(a.ncclAllGather.stream);
+      break;
+    case ROCPROFILER_RCCL_API_ID_ncclReduceScatter:
+      rec.count = a.ncclReduceScatter.recvcount;
+      rec.datatype = a.ncclReduceScatter.datatype;
+      rec.comm = reinterpret_cast<uint64_t>(a.ncclReduceScatter.comm);
+      rec.stream = reinterpret_cast<uint64_t>(a.ncclReduceScatter.stream);
+      break;
+    case ROCPROFILER_RCCL_API_ID_ncclAllToAll:
+      rec.count = a.ncclAllToAll.count;
+      rec.datatype = a.ncclAllToAll.datatype;
+      rec.comm = reinterpret_cast<uint64_t>(a.ncclAllToAll.comm);
+      rec.stream = reinterpret_cast<uint64_t>(a.ncclAllToAll.stream);
+      break;
+    case ROCPROFILER_RCCL_API_ID_ncclBroadcast:
+      rec.count = a.ncclBroadcast.count;
+      rec.datatype = a.ncclBroadcast.datatype;
+      rec.peer_or_root = a.ncclBroadcast.root;
+      rec.comm = reinterpret_cast<uint64_t>(a.ncclBroadcast.comm);
+      rec.stream = reinterpret_cast<uint64_t>(a.ncclBroadcast.stream);
+      break;
+    case ROCPROFILER_RCCL_API_ID_ncclReduce:
+      rec.count = a.ncclReduce.count;
+      rec.datatype = a.ncclReduce.datatype;
+      rec.peer_or_root = a.ncclReduce.root;
+      rec.comm = reinterpret_cast<uint64_t>(a.ncclReduce.comm);
+      rec.stream = reinterpret_cast<uint64_t>(a.ncclReduce.stream);
+      break;
+    case ROCPROFILER_RCCL_API_ID_ncclSend:
+      rec.count = a.ncclSend.count;
+      rec.datatype = a.ncclSend.datatype;
+      rec.peer_or_root = a.ncclSend.peer;
+      rec.comm = reinterpret_cast<uint64_t>(a.ncclSend.comm);
+      rec.stream = reinterpret_cast<uint64_t>(a.ncclSend.stream);
+      break;
+    case ROCPROFILER_RCCL_API_ID_ncclRecv:
+      rec.count = a.ncclRecv.count;
+      rec.datatype = a.ncclRecv.datatype;
+      rec.peer_or_root = a.ncclRecv.peer;
+      rec.comm = reinterpret_cast<uint64_t>(a.ncclRecv.comm);
+      rec.stream = reinterpret_cast<uint64_t>(a.ncclRecv.stream);
+      break;
+    default:
+      break;
+  }
+  rec.elem_size = nccl_elem_size(rec.datatype);
+  write_raw(&rec, sizeof(rec));
+  g_n_records.fetch_add(1, std::memory_order_relaxed);
+}
+
+// roctx markers
+void marker_callback(rocprofiler_callback_tracing_record_t record,
+                     rocprofiler_user_data_t* user_data, void*) {
+  if (record.kind != ROCPROFILER_CALLBACK_TRACING_MARKER_CORE_API) return;
+  auto* data = static_cast<rocprofiler_callback_tracing_marker_api_data_t*>(
+      record.payload);
+  rocprofiler_timestamp_t now = 0;
+  rocprofiler_get_timestamp(&now);
+  if (record.phase == ROCPROFILER_CALLBACK_PHASE_ENTER) {
+    if (record.operation == ROCPROFILER_MARKER_CORE_API_ID_roctxMarkA &&
+        data->args.roctxMarkA.message) {
+      write_name_rec(sgt::REC_MARKER, now, data->args.roctxMarkA.message);
+    } else if (record.operation ==
+                   ROCPROFILER_MARKER_CORE_API_ID_roctxRangePushA &&
+               data->args.roctxRangePushA.message) {
+      write_name_rec(sgt::REC_MARKER, now, data->args.roctxRangePushA.message);
+    }
+  }
+  (void) user_data;
+}
+
+void buffer_callback(rocprofiler_context_id_t, rocprofiler_buffer_id_t,
+                     rocprofiler_record_header_t** headers, size_t num_headers,
+                     void*, uint64_t drop_count) {
+  if (drop_count > 0) {
+    sgt::DropRec d{};
+    d.h = {sgt::REC_DROP, sizeof(sgt::DropRec), 0};
+    d.dropped = drop_count;
+    write_raw(&d, sizeof(d));
+  }
+  // Serialize the batch into one contiguous chunk, then one locked write.
+  std::vector<char> chunk;
+  chunk.reserve(num_headers * 80);
+  auto emit = [&chunk](const void* p, size_t n) {
+    const char* c = static_cast<const char*>(p);
+    chunk.insert(chunk.end(), c, c + n);
+  };
+  for (size_t i = 0; i < num_headers; ++i) {
+    auto* header = headers[i];
+    if (header->category != ROCPROFILER_BUFFER_CATEGORY_TRACING) continue;
+    switch (header->kind) {
+      case ROCPROFILER_BUFFER_TRACING_KERNEL_DISPATCH: {
+        auto* r = static_cast<rocprofiler_buffer_tracing_kernel_dispatch_record_t*>(
+            header->payload);
+        sgt::KernelRec rec{};
+        rec.h = {sgt::REC_KERNEL, sizeof(sgt::KernelRec), 0};
+        rec.start_ns = r->start_timestamp;
+        rec.end_ns = r->end_timestamp;
+        rec.corr_id = r->correlation_id.internal;
+        rec.tid = static_cast<uint32_t>(r->thread_id);
+        rec.device = static_cast<uint32_t>(
+            agent_device(r->dispatch_info.agent_id) < 0
+                ? 0
+                : agent_device(r->dispatch_info.agent_id));
+        rec.queue_id = r->dispatch_info.queue_id.handle;
+        rec.kernel_id = r->dispatch_info.kernel_id;
+        rec.private_segment_size = r->dispatch_info.private_segment_size;
+        rec.group_segment_size = r->dispatch_info.group_segment_size;
+        rec.grid_x = r->dispatch_info.grid_size.x;
+        rec.grid_y = r->dispatch_info.grid_size.y;
+        rec.grid_z = r->dispatch_info.grid_size.z;
+        rec.wg_x = r->dispatch_info.workgroup_size.x;
+        rec.wg_y = r->dispatch_info.workgroup_size.y;
+        rec.wg_z = r->dispatch_info.workgroup_size.z;
+        emit(&rec, sizeof(rec));
+        break;
+      }
+      case ROCPROFILER_BUFFER_TRACING_MEMORY_COPY: {
+        auto* r = static_cast<rocprofiler_buffer_tracing_memory_copy_record_t*>(
+            header->payload);
+        sgt::CopyRec rec{};
+        rec.h = {sgt::REC_COPY, sizeof(sgt::CopyRec), 0};
+        rec.start_ns = r->start_timestamp;
+        rec.end_ns = r->end_timestamp;
+        rec.corr_id = r->correlation_id.internal;
+        rec.tid = static_cast<uint32_t>(r->thread_id);
+        rec.op = r->operation;
+        rec.src_device = agent_device(r->src_agent_id);
+        rec.dst_device = agent_device(r->dst_agent_id);
+        rec.bytes = r->bytes;
+        emit(&rec, sizeof(rec));
+        break;
+      }
+      case ROCPROFILER_BUFFER_TRACING_HIP_RUNTIME_API: {
+        auto* r = static_cast<rocprofiler_buffer_tracing_hip_api_record_t*>(
+            header->payload);
+        sgt::ApiRec rec{};
+        rec.h = {sgt::REC_HIPAPI, sizeof(sgt::ApiRec), 0};
+        rec.start_ns = r->start_timestamp;
+        rec.end_ns = r->end_timestamp;
+        rec.corr_id = r->correlation_id.internal;
+        rec.tid = static_cast<uint32_t>(r->thread_id);
+        rec.op = r->operation;
+        emit(&rec, sizeof(rec));
+        break;
+      }
+      case ROCPROFILER_BUFFER_TRACING_MEMORY_ALLOCATION: {
+        auto* r =
+            static_cast<rocprofiler_buffer_tracing_memory_allocation_record_t*>(
+                header->payload);
+        sgt::AllocRec rec{};
+        rec.h = {sgt::REC_ALLOC, sizeof(sgt::AllocRec), 0};
+        rec.start_ns = r->start_timestamp;
+        rec.end_ns = r->end_timestamp;
+        rec.corr_id = r->correlation_id.internal;
+        rec.tid = static_cast<uint32_t>(r->thread_id);
+        rec.op = r->operation;
+        rec.device = agent_device(r->agent_id);
+        rec.address = r->address.value;
+        rec.bytes = r->allocation_size;
+        emit(&rec, sizeof(rec));
+        break;
+      }
+      default:
+        break;
+    }
+  }
+  if (!chunk.empty()) {
+    write_raw(chunk.data(), chunk.size());
+    g_n_records.fetch_add(num_headers, std::memory_order_relaxed);
+  }
+}
+
+void write_agents() {
+  rocprofiler_query_available_agents(
+      ROCPROFILER_AGENT_INFO_VERSION_0,
+      [](rocprofiler_agent_version_t, const void** agents, size_t num_agents,
+         void*) -> rocprofiler_status_t {
+        for (size_t i = 0; i < num_agents; ++i) {
+          const auto* a =
+              static_cast<const rocprofiler_agent_v0_t*>(agents[i]);
+          int32_t dev = (a->type == ROCPROFILER_AGENT_TYPE_GPU)
+                            ? a->logical_node_type_id
+                            : -1;
+          g_agent_device[a->id.handle] = dev;
+          sgt::AgentRec rec{};
+          rec.h = {sgt::REC_AGENT, sizeof(sgt::AgentRec), 0};
+          rec.agent_handle = a->id.handle;
+          rec.device = dev;
+          rec.type = a->type;
+          rec.node_id = a->node_id;
+          rec.wave_front_size = a->wave_front_size;
+          rec.cu_count = a->cu_count;
+          rec.num_xcc = a->num_xcc;
+          if (a->name) {
+            strncpy(rec.name, a->name, sizeof(rec.name) - 1);
+          }
+          write_raw(&rec, sizeof(rec));
+        }
+        return ROCPROFILER_STATUS_SUCCESS;
+      },
+      sizeof(rocprofiler_agent_v0_t), nullptr);
+}
+
+void write_opnames() {
+  // dump operation-name tables for the kinds we record so preprocess never
+  // needs the SDK at analysis time
+  for (auto kind : {ROCPROFILER_BUFFER_TRACING_HIP_RUNTIME_API,
+                    ROCPROFILER_BUFFER_TRACING_MEMORY_COPY,
+                    ROCPROFILER_BUFFER_TRACING_MEMORY_ALLOCATION,
+                    ROCPROFILER_BUFFER_TRACING_RCCL_API}) {
+    rocprofiler_iterate_buffer_tracing_kind_operations(
+        kind,
+        [](rocprofiler_buffer_tracing_kind_t k, rocprofiler_tracing_operation_t op,
+           void*) -> int {
+          const char* name = nullptr;
+          uint64_t len = 0;
+          if (rocprofiler_query_buffer_tracing_kind_operation_name(
+                  k, op, &name, &len) == ROCPROFILER_STATUS_SUCCESS &&
+              name) {
+            write_opname_rec(k, op, name);
+          }
+          return 1;
+        },
+        nullptr);
+  }
+}
+
+int tool_init(rocprofiler_client_finalize_t, void*) {
+  const char* logdir = getenv("SOFA_LOGDIR");
+  if (!logdir || !*logdir) logdir = ".";
+  char path[4096];
+  snprintf(path, sizeof(path), "%s/gputrace_%d.sgt", logdir, getpid());
+  g_out = fopen(path, "wb");
+  if (!g_out) {
+    fprintf(stderr, "[sofatracer] cannot open %s\n", path);
+    return -1;
+  }
+
+  sgt::FileHeader hdr{};
+  hdr.magic = sgt::kMagic;
+  hdr.version = sgt::kVersion;
+  hdr.pid = static_cast<uint32_t>(getpid());
+  rocprofiler_timestamp_t ts = 0;
+  rocprofiler_get_timestamp(&ts);
+  hdr.realtime_ns = host_ns(CLOCK_REALTIME);
+  hdr.monotonic_raw_ns = host_ns(CLOCK_MONOTONIC_RAW);
+  hdr.rocp_ns = ts;
+  fwrite(&hdr, sizeof(hdr), 1, g_out);
+
+  write_agents();
+  write_opnames();
+  write_clock_rec();
+
+  if (rocprofiler_create_context(&g_ctx) != ROCPROFILER_STATUS_SUCCESS)
+    return -1;
+
+  auto code_object_ops = std::vector<rocprofiler_tracing_operation_t>{
+      ROCPROFILER_CODE_OBJECT_DEVICE_KERNEL_SYMBOL_REGISTER};
+  rocprofiler_configure_callback_tracing_service(
+      g_ctx, ROCPROFILER_CALLBACK_TRACING_CODE_OBJECT, code_object_ops.data(),
+      code_object_ops.size(), code_object_callback, nullptr);
+
+  size_t buffer_mb = 64;
+  if (const char* v = getenv("SOFA_GPU_BUFFER_MB"); v && *v)
+    buffer_mb = strtoull(v, nullptr, 10);
+  size_t buffer_bytes = buffer_mb << 20;
+  if (rocprofiler_create_buffer(g_ctx, buffer_bytes, buffer_bytes / 2,
+                                ROCPROFILER_BUFFER_POLICY_LOSSLESS,
+                                buffer_callback, nullptr,
+                                &g_buffer) != ROCPROFILER_STATUS_SUCCESS)
+    return -1;
+
+  rocprofiler_configure_buffer_tracing_service(
+      g_ctx, ROCPROFILER_BUFFER_TRACING_KERNEL_DISPATCH, nullptr, 0, g_buffer);
+  rocprofiler_configure_buffer_tracing_service(
+      g_ctx, ROCPROFILER_BUFFER_TRACING_MEMORY_COPY, nullptr, 0, g_buffer);
+  if (env_flag("SOFA_TRACE_HIP_API", true)) {
+    rocprofiler_configure_buffer_tracing_service(
+        g_ctx, ROCPROFILER_BUFFER_TRACING_HIP_RUNTIME_API, nullptr, 0,
+        g_buffer);
+  }
+  if (env_flag("SOFA_TRACE_ALLOC", false)) {
+    rocprofiler_configure_buffer_tracing_service(
+        g_ctx, ROCPROFILER_BUFFER_TRACING_MEMORY_ALLOCATION, nullptr, 0,
+        g_buffer);
+  }
+  if (env_flag("SOFA_TRACE_RCCL", true)) {
+    rocprofiler_configure_callback_tracing_service(
+        g_ctx, ROCPROFILER_CALLBACK_TRACING_RCCL_API, nullptr, 0,
+        rccl_callback, nullptr);
+  }
+  rocprofiler_configure_callback_tracing_service(
+      g_ctx, ROCPROFILER_CALLBACK_TRACING_MARKER_CORE_API, nullptr, 0,
+      marker_callback, nullptr);
+
+  auto cb_thread = rocprofiler_callback_thread_t{};
+  if (rocprofiler_create_callback_thread(&cb_thread) ==
+      ROCPROFILER_STATUS_SUCCESS)
+    rocprofiler_assign_callback_thread(g_buffer, cb_thread);
+
+  int valid = 0;
+  rocprofiler_context_is_valid(g_ctx, &valid);
+  if (valid == 0) return -1;
+  rocprofiler_start_context(g_ctx);
+  return 0;
+}
+
+void tool_fini(void*) {
+  rocprofiler_flush_buffer(g_buffer);
+  write_clock_rec();
+  std::lock_guard<std::mutex> lk(g_mutex);
+  if (g_out) {
+    fclose(g_out);
+    g_out = nullptr;
+  }
+}
+
+}  // namespace
+
+extern "C" rocprofiler_tool_configure_result_t*
+rocprofiler_configure(uint32_t version, const char* runtime_version,
+                      uint32_t priority, rocprofiler_client_id_t* id) {
+  id->name = "sofa_amd-tracer";
+  g_client_id = id;
+  (void) version;
+  (void) runtime_version;
+  (void) priority;
+  static auto cfg = rocprofiler_tool_configure_result_t{
+      sizeof(rocprofiler_tool_configure_result_t), &tool_init, &tool_fini,
+      nullptr};
+  return &cfg;
+}

@@ -1,0 +1,230 @@
+// On-device trace ring + compaction kernel — hand-written HIP for CDNA4.
+//
+// The reference pipeline collapses at modern kernel-launch rates because
+// every GPU event takes a CSV->Python-row round trip (SURVEY.md §7 step 6,
+// "event-rate scaling").  The MI355X design keeps a device-resident ring that
+// instrumented GPU code (or future device-side tooling) appends 32-byte
+// records to with one atomic, and a *compaction kernel* that filters +
+// densifies + clock-converts records entirely on device before a single
+// coalesced D2H copy (BASELINE.json north star: "on-device trace-ring
+// compaction kernel ... LDS staging").
+//
+// CDNA4 specifics (see /opt/skills/guides/cdna_hip_programming.md):
+//  * wavefront = 64: the compaction uses 64-bit __ballot + __popcll
+//    prefix-sums (NOT 32-bit warp idioms);
+//  * block-level stable compaction: per-wave totals staged through LDS,
+//    wave 0 scans, one global atomicAdd per block for the output base;
+//  * the tick->ns conversion is fused into the compaction pass (one HBM
+//    round trip, not a separate elementwise kernel);
+//  * records are 32 B so one lane moves one record as 2x b128-class vector
+//    accesses; the output write is dense/coalesced.
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+
+#define HIP_CHECK(x)                                                          \
+  do {                                                                        \
+    hipError_t err_ = (x);                                                    \
+    if (err_ != hipSuccess) {                                                 \
+      fprintf(stderr, "HIP error %s at %s:%d\n", hipGetErrorString(err_),     \
+              __FILE__, __LINE__);                                            \
+      return -1;                                                              \
+    }                                                                         \
+  } while (0)
+
+namespace {
+
+struct __align__(16) RingRec {
+  uint64_t t_start;  // device ticks (s_memrealtime) or pre-converted ns
+  uint64_t t_end;
+  uint32_t tag;      // event class; 0 = empty slot
+  uint32_t src;      // producer id (wave/block/stream tag)
+  uint64_t arg;      // user payload
+};
+static_assert(sizeof(RingRec) == 32, "RingRec must be 32 bytes");
+
+struct RingControl {
+  unsigned long long head;  // total pushes (monotonic)
+  uint32_t capacity;
+  uint32_t _pad;
+};
+
+// ---------------------------------------------------------------- device API
+
+__device__ inline void ring_push(RingControl* ctl, RingRec* slots, uint32_t tag,
+                                 uint32_t src, uint64_t arg, uint64_t t0,
+                                 uint64_t t1) {
+  unsigned long long h = atomicAdd(&ctl->head, 1ull);
+  RingRec r;
+  r.t_start = t0;
+  r.t_end = t1;
+  r.tag = tag;
+  r.src = src;
+  r.arg = arg;
+  slots[h % ctl->capacity] = r;
+}
+
+// Test producer: each thread pushes one deterministic record.  Mirrored by
+// the CPU reference in tests/test_gpu_trace_ring.py.
+__global__ void producer_kernel(RingControl* ctl, RingRec* slots, uint32_t n,
+                                uint32_t n_tags) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  uint64_t t = __builtin_amdgcn_s_memrealtime();
+  ring_push(ctl, slots, 1u + (i % n_tags), i / 64, (uint64_t) i * 3u + 1u, t,
+            t + 100 + (i % 7));
+}
+
+// ------------------------------------------------------- compaction kernel
+//
+// keep records whose tag-class bit is in tag_mask (bit k = tag class k,
+// tag classes are tag % 64); write t' = t * scale + offset (double math —
+// scale is ~10.0 for the 100 MHz counter, exact for the < 2^53 tick values
+// seen in practice).
+
+constexpr int kBlockThreads = 256;          // 4 waves
+constexpr int kWavesPerBlock = kBlockThreads / 64;
+
+__global__ void compact_kernel(const RingRec* __restrict__ slots, uint32_t n_valid,
+                               uint64_t tag_mask, double scale, long long offset,
+                               RingRec* __restrict__ out, uint32_t out_cap,
+                               unsigned int* __restrict__ out_count) {
+  __shared__ uint32_t wave_totals[kWavesPerBlock];
+  __shared__ uint32_t wave_bases[kWavesPerBlock];
+  __shared__ unsigned int block_base;
+
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+
+  RingRec r{};
+  bool keep = false;
+  if (i < n_valid) {
+    r = slots[i];
+    keep = r.tag != 0 && ((tag_mask >> (r.tag & 63u)) & 1ull);
+  }
+
+  // 64-wide ballot compaction (CDNA: __ballot returns a 64-bit mask);
+  // ((1ull<<lane)-1) is well-defined for lane in [0,63]
+  unsigned long long mask = __ballot(keep);
+  uint32_t prefix = __popcll(mask & ((1ull << lane) - 1ull));
+  uint32_t wave_total = __popcll(mask);
+
+  if (lane == 0) wave_totals[wave] = wave_total;
+  __syncthreads();
+
+  if (wave == 0 && lane == 0) {
+    uint32_t run = 0;
+    for (int w = 0; w < kWavesPerBlock; ++w) {
+      wave_bases[w] = run;
+      run += wave_totals[w];
+    }
+    block_base = atomicAdd(out_count, run);
+  }
+  __syncthreads();
+
+  if (keep) {
+    uint32_t dst = block_base + wave_bases[wave] + prefix;
+    if (dst < out_cap) {
+      r.t_start = (uint64_t)((double) r.t_start * scale + (double) offset);
+      r.t_end = (uint64_t)((double) r.t_end * scale + (double) offset);
+      out[dst] = r;
+    }
+  }
+}
+
+struct Ring {
+  RingControl* d_ctl;
+  RingRec* d_slots;
+  uint32_t capacity;
+  int device;
+};
+
+}  // namespace
+
+extern "C" {
+
+int sofa_ring_create(int device, uint32_t capacity, void** ring_out) {
+  HIP_CHECK(hipSetDevice(device));
+  Ring* ring = new Ring{};
+  ring->capacity = capacity;
+  ring->device = device;
+  HIP_CHECK(hipMalloc(&ring->d_ctl, sizeof(RingControl)));
+  HIP_CHECK(hipMalloc(&ring->d_slots, sizeof(RingRec) * (size_t) capacity));
+  RingControl ctl{0, capacity, 0};
+  HIP_CHECK(hipMemcpy(ring->d_ctl, &ctl, sizeof(ctl), hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemset(ring->d_slots, 0, sizeof(RingRec) * (size_t) capacity));
+  *ring_out = ring;
+  return 0;
+}
+
+int sofa_ring_destroy(void* ring_p) {
+  Ring* ring = static_cast<Ring*>(ring_p);
+  HIP_CHECK(hipFree(ring->d_ctl));
+  HIP_CHECK(hipFree(ring->d_slots));
+  delete ring;
+  return 0;
+}
+
+int sofa_ring_test_produce(void* ring_p, uint32_t n, uint32_t n_tags) {
+  Ring* ring = static_cast<Ring*>(ring_p);
+  HIP_CHECK(hipSetDevice(ring->device));
+  dim3 block(256);
+  dim3 grid((n + 255) / 256);
+  hipLaunchKernelGGL(producer_kernel, grid, block, 0, 0, ring->d_ctl,
+                     ring->d_slots, n, n_tags);
+  HIP_CHECK(hipDeviceSynchronize());
+  return 0;
+}
+
+int sofa_ring_head(void* ring_p, unsigned long long* head_out) {
+  Ring* ring = static_cast<Ring*>(ring_p);
+  RingControl ctl;
+  HIP_CHECK(hipMemcpy(&ctl, ring->d_ctl, sizeof(ctl), hipMemcpyDeviceToHost));
+  *head_out = ctl.head;
+  return 0;
+}
+
+// Compact + convert + copy back.  host_out must hold max_out records.
+// n_out gets the number of kept records (clamped to max_out on copy).
+int sofa_ring_compact(void* ring_p, uint64_t tag_mask, double scale,
+                      long long offset, void* host_out, uint32_t max_out,
+                      uint32_t* n_out) {
+  Ring* ring = static_cast<Ring*>(ring_p);
+  HIP_CHECK(hipSetDevice(ring->device));
+  RingControl ctl;
+  HIP_CHECK(hipMemcpy(&ctl, ring->d_ctl, sizeof(ctl), hipMemcpyDeviceToHost));
+  uint32_t n_valid =
+      (uint32_t)(ctl.head < ctl.capacity ? ctl.head : ctl.capacity);
+
+  RingRec* d_out = nullptr;
+  unsigned int* d_count = nullptr;
+  HIP_CHECK(hipMalloc(&d_out, sizeof(RingRec) * (size_t) n_valid));
+  HIP_CHECK(hipMalloc(&d_count, sizeof(unsigned int)));
+  HIP_CHECK(hipMemset(d_count, 0, sizeof(unsigned int)));
+
+  if (n_valid > 0) {
+    dim3 block(kBlockThreads);
+    dim3 grid((n_valid + kBlockThreads - 1) / kBlockThreads);
+    hipLaunchKernelGGL(compact_kernel, grid, block, 0, 0, ring->d_slots,
+                       n_valid, tag_mask, scale, offset, d_out, n_valid,
+                       d_count);
+    HIP_CHECK(hipDeviceSynchronize());
+  }
+
+  unsigned int count = 0;
+  HIP_CHECK(hipMemcpy(&count, d_count, sizeof(count), hipMemcpyDeviceToHost));
+  uint32_t n_copy = count < max_out ? count : max_out;
+  if (n_copy > 0) {
+    HIP_CHECK(hipMemcpy(host_out, d_out, sizeof(RingRec) * (size_t) n_copy,
+                        hipMemcpyDeviceToHost));
+  }
+  HIP_CHECK(hipFree(d_out));
+  HIP_CHECK(hipFree(d_count));
+  *n_out = count;
+  return 0;
+}
+
+}  // extern "C"

@@ -1,0 +1,269 @@
+"""sofa record — orchestration.
+
+Behavioral parity with reference bin/sofa_record.py:150-523 (preflight,
+prologue clock sync, background monitors, target launch, epilogue), rebuilt
+for ROCm:
+
+* GPU activity: rocprofiler-sdk collector library injected via
+  ROCP_TOOL_LIBRARIES (replaces the system-wide nvprof daemon,
+  bin/sofa_record.py:217-221).
+* CPU samples: native sofa-cpusampler on perf_event_open (replaces
+  `perf record -F 99`, :339-354).
+* Clock sync: every collector stamps CLOCK_MONOTONIC_RAW + CLOCK_REALTIME
+  directly (replaces the sofa_perf_timebase + cuhello pairing,
+  :236-242; see native/timebase/timebase.cc).
+* Telemetry: one SysMonitor thread + librocm_smi64 ctypes
+  (replaces ~10 monitor subprocesses, :249-312).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import shutil
+import signal
+import subprocess
+import sys
+import time
+
+from ..config import SofaConfig, ensure_logdir
+from .. import printing as p
+from .pollers import SysMonitor, dump_xgmi_topology
+
+RAW_FILES = [
+    "cpusamples.scs",
+    "timebase.json",
+    "sofa_time.txt",
+    "cpuinfo.txt",
+    "mpstat.txt",
+    "diskstat.txt",
+    "netstat.txt",
+    "vmstat.txt",
+    "gpusmi.txt",
+    "xgmi_topo.txt",
+    "kallsyms",
+    "misc.txt",
+    "pktcap.bin",
+    "sofa.err",
+]
+
+DERIVED_FILES = [
+    "cputrace.csv",
+    "gputrace.csv",
+    "rccltrace.csv",
+    "comm.csv",
+    "mpstat.csv",
+    "usr_sys.csv",
+    "diskstat.csv",
+    "diskstat_vector.csv",
+    "diskstat_vector_ui.csv",
+    "vmstat.csv",
+    "netstat.csv",
+    "netbandwidth.csv",
+    "nettrace.csv",
+    "gpusmi_trace.csv",
+    "hip_api_trace.csv",
+    "report.js",
+    "performance.csv",
+    "netrank.csv",
+    "auto_caption.csv",
+    "swarms_report.txt",
+    "iteration_timeline.txt",
+    "features.csv",
+    "xlink_traffic.csv",
+]
+
+
+def native_dir() -> str:
+    return os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "native")
+
+
+def native_bin(name: str) -> str:
+    return os.path.join(native_dir(), "bin", name)
+
+
+def native_lib(name: str) -> str:
+    return os.path.join(native_dir(), "lib", name)
+
+
+def ensure_native_built(verbose: bool = False) -> bool:
+    """Build native helpers on demand (first run on a new machine)."""
+    needed = [native_bin("sofa-cpusampler"), native_bin("sofa-timebase"), native_lib("libsofatracer.so")]
+    if all(os.path.exists(x) for x in needed):
+        return True
+    try:
+        from ..native.build import build_all
+
+        build_all(verbose=verbose)
+    except Exception as e:  # pragma: no cover
+        p.print_warning(f"native build failed: {e}")
+    return all(os.path.exists(x) for x in needed)
+
+
+def sofa_clean(cfg: SofaConfig) -> None:
+    """Delete raw + derived artifacts (reference bin/sofa_record.py:138-147)."""
+    logdir = cfg.logdir
+    if not os.path.isdir(logdir):
+        return
+    for name in os.listdir(logdir):
+        full = os.path.join(logdir, name)
+        if (
+            name in RAW_FILES
+            or name in DERIVED_FILES
+            or name.startswith(("gputrace_", "sofa_hints"))
+            or name.endswith((".sgt", ".scs"))
+        ):
+            if os.path.isdir(full):
+                shutil.rmtree(full, ignore_errors=True)
+            else:
+                try:
+                    os.remove(full)
+                except OSError:
+                    pass
+    p.print_progress(f"cleaned {logdir}")
+
+
+def _write_timebase(logdir: str) -> None:
+    tb_bin = native_bin("sofa-timebase")
+    out = os.path.join(logdir, "timebase.json")
+    try:
+        with open(out, "w") as f:
+            subprocess.run([tb_bin, "3"], stdout=f, check=True, timeout=10)
+    except (OSError, subprocess.SubprocessError):
+        # fallback: do it in python (time.clock_gettime has every clock we need)
+        with open(out, "w") as f:
+            for _ in range(3):
+                f.write(
+                    json.dumps(
+                        {
+                            "realtime_ns": time.clock_gettime_ns(time.CLOCK_REALTIME),
+                            "monotonic_ns": time.clock_gettime_ns(time.CLOCK_MONOTONIC),
+                            "monotonic_raw_ns": time.clock_gettime_ns(time.CLOCK_MONOTONIC_RAW),
+                            "boottime_ns": time.clock_gettime_ns(time.CLOCK_BOOTTIME),
+                        }
+                    )
+                    + "\n"
+                )
+
+
+def build_target_env(cfg: SofaConfig) -> dict:
+    env = dict(os.environ)
+    if cfg.enable_gpu:
+        tracer = native_lib("libsofatracer.so")
+        if os.path.exists(tracer):
+            prev = env.get("ROCP_TOOL_LIBRARIES", "")
+            env["ROCP_TOOL_LIBRARIES"] = tracer + ((":" + prev) if prev else "")
+            env["SOFA_LOGDIR"] = os.path.abspath(cfg.logdir)
+            env["SOFA_TRACE_HIP_API"] = "1" if cfg.enable_gpu_hip_api else "0"
+            env["SOFA_TRACE_RCCL"] = "1" if cfg.enable_rccl_trace else "0"
+            env["SOFA_GPU_BUFFER_MB"] = str(cfg.gpu_ring_buffer_mb)
+        else:
+            p.print_warning("libsofatracer.so not built; GPU tracing disabled")
+    return env
+
+
+def sofa_record(command: str, cfg: SofaConfig) -> int:
+    logdir = ensure_logdir(cfg)
+    ensure_native_built(cfg.verbose)
+    sofa_clean(cfg)
+
+    p.print_progress(f"recording into {logdir}: {command}")
+
+    # --- prologue: clock base + symbols + topology ---
+    with open(os.path.join(logdir, "sofa_time.txt"), "w") as f:
+        f.write("%.9f\n" % time.time())
+    _write_timebase(logdir)
+    try:
+        shutil.copyfile("/proc/kallsyms", os.path.join(logdir, "kallsyms"))
+    except OSError:
+        pass
+    if cfg.enable_gpu:
+        dump_xgmi_topology(logdir)
+
+    # --- background monitors ---
+    mon = SysMonitor(logdir, rate_hz=cfg.sys_mon_rate, enable_gpu=cfg.enable_gpu)
+    mon.start()
+
+    pktcap_proc = None
+    if cfg.enable_tcpdump:
+        sniffer = native_bin("sofa-pktcap")
+        if os.path.exists(sniffer):
+            try:
+                pktcap_proc = subprocess.Popen(
+                    [sniffer, "-o", os.path.join(logdir, "pktcap.bin")],
+                    stderr=subprocess.DEVNULL,
+                )
+            except OSError:
+                pktcap_proc = None
+        else:
+            p.print_warning("sofa-pktcap not built; packet capture disabled")
+
+    # --- launch target ---
+    env = build_target_env(cfg)
+    err_f = open(os.path.join(logdir, "sofa.err"), "w")
+    t_begin = time.time()
+    target = subprocess.Popen(
+        ["bash", "-c", command],
+        env=env,
+        stdout=None,
+        stderr=err_f if not cfg.verbose else None,
+        start_new_session=False,
+    )
+
+    # --- CPU sampler attached to the target ---
+    sampler = None
+    sampler_bin = native_bin("sofa-cpusampler")
+    if os.path.exists(sampler_bin):
+        args = [sampler_bin, "-o", os.path.join(logdir, "cpusamples.scs"), "-F", str(cfg.cpu_sample_rate)]
+        if cfg.profile_all_cpus:
+            args.append("-a")
+        else:
+            args += ["-p", str(target.pid)]
+        try:
+            sampler = subprocess.Popen(args)
+        except OSError as e:
+            p.print_warning(f"cpusampler failed to start: {e}")
+    else:
+        p.print_warning("sofa-cpusampler not built; CPU sampling disabled")
+
+    # --- wait ---
+    try:
+        ret = target.wait()
+    except KeyboardInterrupt:
+        target.send_signal(signal.SIGINT)
+        ret = target.wait()
+    t_end = time.time()
+
+    # --- epilogue ---
+    if sampler is not None:
+        try:
+            sampler.terminate()
+            sampler.wait(timeout=5)
+        except (OSError, subprocess.TimeoutExpired):
+            sampler.kill()
+    if pktcap_proc is not None:
+        try:
+            pktcap_proc.terminate()
+            pktcap_proc.wait(timeout=5)
+        except (OSError, subprocess.TimeoutExpired):
+            pktcap_proc.kill()
+    mon.stop()
+    mon.join(timeout=5)
+    err_f.close()
+
+    with open(os.path.join(logdir, "misc.txt"), "w") as f:
+        f.write(
+            json.dumps(
+                {
+                    "elapsed_time": t_end - t_begin,
+                    "cores": os.cpu_count(),
+                    "pid": target.pid,
+                    "returncode": ret,
+                    "command": command,
+                }
+            )
+        )
+    if ret != 0:
+        p.print_warning(f"target exited with {ret}")
+    p.print_progress("recording done (elapsed %.2f s)" % (t_end - t_begin))
+    return ret
