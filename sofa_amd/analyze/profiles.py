@@ -1,0 +1,217 @@
+"""Per-subsystem profiles, each appending (name, value) features.
+
+Parity map (reference bin/sofa_analyze.py):
+  cpu_profile :694-710, gpu_profile :343-377, nvsmi_profile :259-341,
+  mpstat_profile :735-790, vmstat_profile :712-733, diskstat_profile
+  :640-692, netbandwidth_profile :531-594, net_profile :385-493,
+  spotlight ROI :873-896.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional, Tuple
+
+import numpy as np
+import pandas as pd
+
+from .. import printing as p
+
+Features = List[Tuple[str, float]]
+
+
+def _q(series) -> Tuple[float, float, float, float]:
+    a = np.asarray(series, dtype=np.float64)
+    if len(a) == 0:
+        return (0.0, 0.0, 0.0, 0.0)
+    return tuple(np.percentile(a, [25, 50, 75, 100]))
+
+
+def cpu_profile(df_cpu: pd.DataFrame, features: Features, elapsed: float) -> None:
+    if df_cpu is None or len(df_cpu) == 0:
+        return
+    total = df_cpu["duration"].sum()
+    features.append(("cpu_time", float(total)))
+    features.append(("n_cpu_samples", float(len(df_cpu))))
+    print("\nCPU profile:")
+    print("  sampled cpu time: %.3f s over %d samples" % (total, len(df_cpu)))
+    top = (
+        df_cpu.groupby("name")["duration"]
+        .agg(["sum", "count"])
+        .sort_values("sum", ascending=False)
+        .head(10)
+    )
+    print("  top functions:")
+    for name, row in top.iterrows():
+        short = name if len(str(name)) < 90 else str(name)[:87] + "..."
+        print("    %8.3f s %6d  %s" % (row["sum"], int(row["count"]), short))
+
+
+def gpu_profile(df_gpu: pd.DataFrame, df_rccl: pd.DataFrame, features: Features) -> None:
+    if df_gpu is None or len(df_gpu) == 0:
+        return
+    print("\nGPU profile:")
+    kernels = df_gpu[df_gpu["copyKind"] == 0]
+    copies = df_gpu[df_gpu["copyKind"] != 0]
+    gpu_time = df_gpu["duration"].sum()
+    features.append(("gpu_time", float(gpu_time)))
+    features.append(("gpu_kernel_time", float(kernels["duration"].sum())))
+    features.append(("gpu_memcpy_time", float(copies["duration"].sum())))
+    features.append(("n_gpu_events", float(len(df_gpu))))
+    # rccl time: prefer the API trace, fall back to kernel-name grep
+    # (reference greps `nccl` in kernel names, bin/sofa_analyze.py:363-368)
+    if df_rccl is not None and len(df_rccl):
+        rccl_time = df_rccl["duration"].sum()
+    else:
+        names = kernels["name"].astype(str)
+        rccl_time = kernels[names.str.contains("rccl|nccl|Ccl", case=False, regex=True)][
+            "duration"
+        ].sum()
+    features.append(("rccl_kernel_time", float(rccl_time)))
+    for dev, grp in df_gpu.groupby("deviceId"):
+        print(
+            "  gpu%-2d  time %8.4f s  kernels %6d  copies %6d"
+            % (
+                dev,
+                grp["duration"].sum(),
+                (grp["copyKind"] == 0).sum(),
+                (grp["copyKind"] != 0).sum(),
+            )
+        )
+    top = (
+        kernels.groupby("name")["duration"]
+        .agg(["sum", "count"])
+        .sort_values("sum", ascending=False)
+        .head(10)
+    )
+    if len(top):
+        print("  top kernels:")
+        for name, row in top.iterrows():
+            short = str(name) if len(str(name)) < 90 else str(name)[:87] + "..."
+            print("    %8.4f s %6d  %s" % (row["sum"], int(row["count"]), short))
+
+
+def gpusmi_profile(df_sm: pd.DataFrame, features: Features) -> None:
+    if df_sm is None or len(df_sm) == 0:
+        return
+    print("\nGPU utilization (rocm-smi):")
+    for dev, grp in df_sm.groupby("deviceId"):
+        q25, q50, q75, q100 = _q(grp["duration"])
+        print("  gpu%-2d busy%%: q25=%.0f q50=%.0f q75=%.0f max=%.0f" % (dev, q25, q50, q75, q100))
+        features.append((f"gpu{dev}_util_q50", float(q50)))
+        features.append((f"gpu{dev}_util_max", float(q100)))
+
+
+def mpstat_profile(df_mp: pd.DataFrame, features: Features, idle_threshold: float = 10.0) -> None:
+    if df_mp is None or len(df_mp) == 0:
+        return
+    print("\nCPU cores (mpstat):")
+    busy_by_core = df_mp.groupby("deviceId")["duration"].mean()
+    active = (busy_by_core > idle_threshold).sum()
+    features.append(("cpu_active_ratio", float(active) / max(len(busy_by_core), 1)))
+    features.append(("cpu_mean_busy", float(busy_by_core.mean())))
+    print(
+        "  %d/%d cores active (>%.0f%% busy); mean busy %.1f%%"
+        % (active, len(busy_by_core), idle_threshold, busy_by_core.mean())
+    )
+
+
+def vmstat_profile(logdir: str, features: Features) -> None:
+    path = os.path.join(logdir, "vmstat.csv")
+    if not os.path.isfile(path):
+        return
+    try:
+        d = pd.read_csv(path)
+    except (OSError, ValueError):
+        return
+    if len(d) == 0:
+        return
+    print("\nVM stats:")
+    for col, feat in [("ctxt_r", "ctxt_per_s"), ("pgpgin_r", "pgin_per_s"), ("pgpgout_r", "pgout_per_s")]:
+        if col in d:
+            features.append((feat, float(d[col].mean())))
+    print(
+        "  ctx-switch/s mean %.0f; pgin/s %.0f; pgout/s %.0f"
+        % (d.get("ctxt_r", pd.Series([0])).mean(), d.get("pgpgin_r", pd.Series([0])).mean(), d.get("pgpgout_r", pd.Series([0])).mean())
+    )
+
+
+def diskstat_profile(logdir: str, features: Features) -> None:
+    path = os.path.join(logdir, "diskstat_vector.csv")
+    if not os.path.isfile(path):
+        return
+    try:
+        d = pd.read_csv(path)
+    except (OSError, ValueError):
+        return
+    if len(d) == 0:
+        return
+    print("\nDisk profile:")
+    for dev, grp in d.groupby("dev"):
+        rb = grp["read_Bps"].mean() / 1e6
+        wb = grp["write_Bps"].mean() / 1e6
+        if rb + wb < 0.01:
+            continue
+        print(
+            "  %-10s read %8.1f MB/s  write %8.1f MB/s  r_await %6.2f ms  w_await %6.2f ms"
+            % (dev, rb, wb, grp["r_await_ms"].mean(), grp["w_await_ms"].mean())
+        )
+    features.append(("disk_read_Bps", float(d["read_Bps"].mean())))
+    features.append(("disk_write_Bps", float(d["write_Bps"].mean())))
+
+
+def netbandwidth_profile(logdir: str, features: Features) -> None:
+    path = os.path.join(logdir, "netbandwidth.csv")
+    if not os.path.isfile(path):
+        return
+    try:
+        d = pd.read_csv(path)
+    except (OSError, ValueError):
+        return
+    if len(d) == 0:
+        return
+    print("\nNetwork bandwidth:")
+    for col, feat in [("rx_Bps", "net_rx"), ("tx_Bps", "net_tx")]:
+        q25, q50, q75, q100 = _q(d[col])
+        print("  %s: q25=%.2f q50=%.2f q75=%.2f max=%.2f MB/s" % (feat, q25 / 1e6, q50 / 1e6, q75 / 1e6, q100 / 1e6))
+        features.append((feat + "_q50", float(q50)))
+        features.append((feat + "_max", float(q100)))
+
+
+def net_profile(logdir: str, df_net: pd.DataFrame, features: Features) -> None:
+    """Packet src->dst ranking -> netrank.csv (reference :385-493)."""
+    if df_net is None or len(df_net) == 0:
+        return
+    g = (
+        df_net.groupby(["pkt_src", "pkt_dst"])
+        .agg(packets=("payload", "count"), bytes=("payload", "sum"))
+        .reset_index()
+        .sort_values("bytes", ascending=False)
+    )
+    g.to_csv(os.path.join(logdir, "netrank.csv"), index=False)
+    features.append(("net_num_peers", float(len(g))))
+    print("\nNetwork peers (top 5 by bytes):")
+    print(g.head(5).to_string(index=False))
+
+
+def spotlight_roi(df_sm: pd.DataFrame, up: float = 50.0, down: float = 10.0, trigger: int = 10):
+    """Hysteresis ROI over GPU busy% (reference :873-896).
+
+    Counts up while busy >= `up`, down while busy < `down`; the first window
+    whose count reaches `trigger` opens the ROI, the last closes it.
+    """
+    if df_sm is None or len(df_sm) == 0:
+        return (0.0, 0.0)
+    d = df_sm.sort_values("timestamp")
+    count = 0
+    begin = end = 0.0
+    for ts, util in zip(d["timestamp"], d["duration"]):
+        if util >= up:
+            count += 1
+            if count >= trigger and begin == 0.0:
+                begin = ts
+            if begin > 0.0:
+                end = ts
+        elif util < down:
+            count = max(0, count - 1)
+    return (begin, end)

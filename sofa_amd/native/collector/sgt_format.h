@@ -59,7 +59,7 @@ struct KernelRec {
   uint32_t grid_x, grid_y, grid_z;
   uint32_t wg_x, wg_y, wg_z;
   uint32_t pad2;
-};  // 80 bytes
+};  // 96 bytes
 
 struct CopyRec {
   RecHeader h;        // REC_COPY
@@ -71,7 +71,7 @@ struct CopyRec {
   int32_t src_device; // logical GPU index or -1 for host
   int32_t dst_device;
   uint64_t bytes;
-};  // 48 bytes
+};  // 56 bytes
 
 struct ApiRec {
   RecHeader h;        // REC_HIPAPI
@@ -96,7 +96,7 @@ struct RcclRec {
   uint32_t device;    // current HIP device at call time
   uint64_t comm;      // ncclComm_t pointer value (communicator identity)
   uint64_t stream;    // hipStream_t pointer value
-};  // 72 bytes
+};  // 80 bytes
 
 struct NameRec {  // REC_KERNEL_NAME / REC_MARKER: header + id + chars
   RecHeader h;
@@ -121,7 +121,7 @@ struct AgentRec {
   uint32_t cu_count;
   uint32_t num_xcc;
   char name[64];        // gfx name, NUL-terminated
-};  // 96 bytes
+};  // 104 bytes
 
 struct ClockRec {
   RecHeader h;  // REC_CLOCK

@@ -214,8 +214,8 @@ void drain_ring(Ring& r, Writer& w) {
         size_t flen = strnlen(fname, fmax);
         // executable mappings only (reduce noise): prot & PROT_EXEC
         if (!(mm->prot & 4)) break;
-        size_t total = sizeof(MmapRec) + flen + 1;
-        std::vector<char> out(total);
+        size_t total = (sizeof(MmapRec) + flen + 1 + 7) & ~size_t(7);
+        std::vector<char> out(total, 0);
         auto* rec = reinterpret_cast<MmapRec*>(out.data());
         rec->h = {REC_MMAP, static_cast<uint16_t>(total)};
         rec->time_ns = 0;  // mmap records carry no time with our sample_id_all=0
@@ -237,8 +237,8 @@ void drain_ring(Ring& r, Writer& w) {
         const char* comm = body + sizeof(C);
         size_t cmax = esz - sizeof(struct perf_event_header) - sizeof(C);
         size_t clen = strnlen(comm, cmax);
-        size_t total = sizeof(CommRec) + clen + 1;
-        std::vector<char> out(total);
+        size_t total = (sizeof(CommRec) + clen + 1 + 7) & ~size_t(7);
+        std::vector<char> out(total, 0);
         auto* rec = reinterpret_cast<CommRec*>(out.data());
         rec->h = {REC_COMM, static_cast<uint16_t>(total)};
         rec->time_ns = 0;
@@ -386,8 +386,8 @@ int main(int argc, char** argv) {
                      &lo, &hi, perms, &off, fname);
       if (n < 4 || strchr(perms, 'x') == nullptr || fname[0] != '/') continue;
       size_t flen = strlen(fname);
-      size_t total = sizeof(MmapRec) + flen + 1;
-      std::vector<char> out(total);
+      size_t total = (sizeof(MmapRec) + flen + 1 + 7) & ~size_t(7);
+      std::vector<char> out(total, 0);
       auto* rec = reinterpret_cast<MmapRec*>(out.data());
       rec->h = {REC_MMAP, static_cast<uint16_t>(total)};
       rec->time_ns = 0;

@@ -1,0 +1,81 @@
+"""CPU samples -> cputrace.csv (unified schema).
+
+Schema semantics follow the reference (bin/sofa_preprocess.py:110-154):
+``event`` = log10(instruction pointer), ``duration`` = sampled CPU time in
+seconds (here: the cpu-clock sample period, exact, instead of the reference's
+cycles/MHz estimate).  Names are "symbol @ dso" resolved offline
+(symbols.Symbolizer), demangled for the viz copy (:1814-1816).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import numpy as np
+import pandas as pd
+
+from ..config import SofaConfig
+from ..schema import new_trace_df
+from .scs import ScsFile, parse_scs
+from .symbols import Symbolizer
+from .timebase import TimeBase
+
+
+def load_scs(logdir: str) -> Optional[ScsFile]:
+    path = os.path.join(logdir, "cpusamples.scs")
+    if not os.path.isfile(path):
+        return None
+    return parse_scs(path)
+
+
+def scs_to_cputrace(
+    scs: ScsFile, tb: Optional[TimeBase], logdir: str = "", symbolize: bool = True
+) -> pd.DataFrame:
+    s = scs.samples
+    n = len(s)
+    df = new_trace_df(n)
+    if n == 0:
+        return df
+    # sampler stamps CLOCK_MONOTONIC_RAW (cpusampler.cc attr.use_clockid)
+    mono = s["time_ns"].astype(np.int64)
+    if tb is not None:
+        # the scs header carries its own clock pair; prefer it (same process)
+        off = scs.realtime_ns - scs.monotonic_raw_ns
+        ts = ((mono + off) * 1e-9) - tb.time_base
+    else:
+        ts = mono * 1e-9
+    df["timestamp"] = ts
+    ips = s["ip"].astype(np.float64)
+    with np.errstate(divide="ignore"):
+        df["event"] = np.where(ips > 0, np.log10(np.maximum(ips, 1.0)), 0.0)
+    df["duration"] = s["period"].astype(np.float64) * 1e-9
+    df["deviceId"] = s["cpu"].astype(np.int64)
+    df["pid"] = s["pid"].astype(np.int64)
+    df["tid"] = s["tid"].astype(np.int64)
+    df["category"] = 0
+
+    if symbolize:
+        kallsyms = os.path.join(logdir, "kallsyms") if logdir else ""
+        if kallsyms and not os.path.isfile(kallsyms):
+            kallsyms = ""
+        symr = Symbolizer(scs.mmaps, kallsyms)
+        # resolve unique (pid, ip, kernel-flag) triples only
+        kern = (s["flags"] & 1).astype(bool)
+        keys = {}
+        names = [""] * n
+        for i in range(n):
+            key = (int(s["pid"][i]), int(s["ip"][i]), bool(kern[i]))
+            nm = keys.get(key)
+            if nm is None:
+                sym, dso = symr.resolve(key[0], key[1], key[2])
+                comm = scs.comms.get(int(s["tid"][i]), "")
+                nm = f"{sym} @ {dso}"
+                if comm:
+                    nm = f"{nm} [{comm}]"
+                keys[key] = nm
+            names[i] = nm
+        df["name"] = names
+    else:
+        df["name"] = ["ip_0x%x" % ip for ip in s["ip"]]
+    return df

@@ -1,0 +1,186 @@
+"""sofa preprocess — raw logdir -> unified CSVs + report.js.
+
+Orchestration parity with reference bin/sofa_preprocess.py:377-2104, rebuilt
+on binary collectors + vectorized parsers.  Every stream is optional: a
+missing raw file only warns (the reference's tested no-GPU degradation path,
+SURVEY.md §4).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional
+
+import numpy as np
+import pandas as pd
+
+from .. import printing as p
+from ..config import SofaConfig
+from ..schema import SOFATrace, new_trace_df, traces_to_json, write_trace_csv
+from . import cpu as cpu_mod
+from . import gpu as gpu_mod
+from . import sysmon
+from .timebase import TimeBase, load_timebase
+
+FILTER_COLORS = [
+    "red", "orange", "yellow", "green", "blue", "indigo", "violet",
+    "cyan", "magenta", "brown",
+]
+
+
+def _filter_traces(df: pd.DataFrame, filters, prefix: str) -> List[SOFATrace]:
+    """Per-keyword colored sub-series (reference gpu/cpu_filters behavior,
+    bin/sofa_preprocess.py:1439-1452,1818-1827)."""
+    out = []
+    if df is None or len(df) == 0:
+        return out
+    names = df["name"].astype(str)
+    for i, f in enumerate(filters or []):
+        sel = df[names.str.contains(f.keyword, case=False, regex=False)]
+        if len(sel) == 0:
+            continue
+        out.append(
+            SOFATrace(
+                name=f"{prefix}_filter_{i}",
+                title=f"{prefix}:{f.keyword}",
+                color=f.color or FILTER_COLORS[i % len(FILTER_COLORS)],
+                data=sel,
+            )
+        )
+    return out
+
+
+def sofa_preprocess(cfg: SofaConfig) -> dict:
+    logdir = cfg.logdir
+    if not os.path.isdir(logdir):
+        p.print_error(f"logdir {logdir} does not exist — run `sofa record` first")
+        return {}
+    p.print_progress(f"preprocessing {logdir}")
+    tb = load_timebase(logdir, cfg.cpu_time_offset_ms)
+    traces: List[SOFATrace] = []
+    result = {"tb": tb}
+
+    # ---------------- CPU samples ----------------
+    df_cpu = new_trace_df(0)
+    scs = None
+    try:
+        scs = cpu_mod.load_scs(logdir)
+    except (ValueError, OSError) as e:
+        p.print_warning(f"cpusamples.scs unreadable: {e}")
+    if scs is not None:
+        df_cpu = cpu_mod.scs_to_cputrace(scs, tb, logdir=logdir, symbolize=True)
+        if scs.lost:
+            p.print_warning(f"cpu sampler lost {scs.lost} records")
+    if len(df_cpu):
+        write_trace_csv(df_cpu, os.path.join(logdir, "cputrace.csv"))
+        traces.append(
+            SOFATrace(name="cpu_traces", title="CPU samples", color="DarkGray", data=df_cpu)
+        )
+        traces += _filter_traces(df_cpu, cfg.cpu_filters, "cpu")
+        # optional swarm clustering (reference --enable_swarms,
+        # bin/sofa_preprocess.py:1828-1836)
+        if cfg.enable_swarms:
+            try:
+                from ..ml.hsg import hsg_cluster, swarms_to_traces
+
+                swarms, captions = hsg_cluster(df_cpu, cfg.num_swarms, logdir)
+                traces += swarms_to_traces(swarms, captions, logdir)
+            except Exception as e:
+                p.print_warning(f"swarm clustering failed: {e}")
+    else:
+        p.print_warning("no CPU samples")
+    result["df_cpu"] = df_cpu
+
+    # ---------------- system monitors ----------------
+    t_mp, mp_csv, usr_sys = sysmon.parse_mpstat(logdir, tb)
+    if len(t_mp):
+        traces.append(SOFATrace(name="mpstat_traces", title="CPU core busy (%)", color="CadetBlue", data=t_mp))
+        mp_csv.to_csv(os.path.join(logdir, "mpstat.csv"), index=False)
+        usr_sys.to_csv(os.path.join(logdir, "usr_sys.csv"), index=False)
+    result["df_mpstat"] = t_mp
+
+    t_disk, disk_vec = sysmon.parse_diskstat(logdir, tb)
+    if len(t_disk):
+        traces.append(SOFATrace(name="diskstat_traces", title="Disk throughput (MB/s)", color="SandyBrown", data=t_disk))
+        disk_vec.to_csv(os.path.join(logdir, "diskstat_vector.csv"), index=False)
+        # UI view: the most active device, or --diskstat_filters selection
+        sel_dev = None
+        if cfg.diskstat_filters:
+            kw = cfg.diskstat_filters[0].keyword
+            if (disk_vec["dev"] == kw).any():
+                sel_dev = kw
+        if sel_dev is None:
+            totals = disk_vec.groupby("dev")[["read_Bps", "write_Bps"]].sum().sum(axis=1)
+            sel_dev = totals.idxmax()
+        disk_vec[disk_vec["dev"] == sel_dev].to_csv(
+            os.path.join(logdir, "diskstat_vector_ui.csv"), index=False
+        )
+    result["df_diskstat"] = t_disk
+
+    t_net, net_bw = sysmon.parse_netstat(logdir, tb)
+    if len(t_net):
+        traces.append(SOFATrace(name="netstat_traces", title="NIC throughput (MB/s)", color="YellowGreen", data=t_net))
+        net_bw.to_csv(os.path.join(logdir, "netbandwidth.csv"), index=False)
+        net_bw.to_csv(os.path.join(logdir, "netstat.csv"), index=False)
+    result["df_netstat"] = t_net
+
+    t_vm, vm_csv = sysmon.parse_vmstat(logdir, tb)
+    if len(t_vm):
+        traces.append(SOFATrace(name="vmstat_traces", title="Context switches/s", color="LightSteelBlue", data=t_vm))
+        vm_csv.to_csv(os.path.join(logdir, "vmstat.csv"), index=False)
+    result["df_vmstat"] = t_vm
+
+    t_sm, t_gmem, gpusmi_csv = sysmon.parse_gpusmi(logdir, tb)
+    if len(t_sm):
+        traces.append(SOFATrace(name="gpusmi_sm_traces", title="GPU busy (%)", color="DarkOrange", data=t_sm))
+        traces.append(SOFATrace(name="gpusmi_mem_traces", title="GPU mem busy (%)", color="Gold", data=t_gmem))
+        gpusmi_csv.to_csv(os.path.join(logdir, "gpusmi_trace.csv"), index=False)
+    result["df_gpusmi"] = t_sm
+
+    # ---------------- GPU activity ----------------
+    df_gpu = new_trace_df(0)
+    df_rccl = new_trace_df(0)
+    df_hip = new_trace_df(0)
+    sgt_files = gpu_mod.load_sgt_files(logdir)
+    if sgt_files:
+        df_gpu = gpu_mod.sgt_to_gputrace(sgt_files, tb)
+        df_rccl = gpu_mod.sgt_to_rccltrace(sgt_files, tb)
+        df_hip = gpu_mod.sgt_to_hip_api_trace(sgt_files, tb)
+        dropped = sum(s.dropped for s in sgt_files)
+        if dropped:
+            p.print_warning(f"GPU collector dropped {dropped} records")
+    if len(df_gpu):
+        write_trace_csv(df_gpu, os.path.join(logdir, "gputrace.csv"))
+        traces.append(SOFATrace(name="gpu_traces", title="GPU kernels & copies", color="DarkSlateBlue", data=df_gpu))
+        traces += _filter_traces(df_gpu, cfg.gpu_filters, "gpu")
+    if len(df_rccl):
+        write_trace_csv(df_rccl, os.path.join(logdir, "rccltrace.csv"))
+        traces.append(SOFATrace(name="rccl_traces", title="RCCL collectives", color="Crimson", data=df_rccl))
+    if len(df_hip) and cfg.enable_gpu_hip_api:
+        write_trace_csv(df_hip, os.path.join(logdir, "hip_api_trace.csv"))
+        traces.append(SOFATrace(name="hip_api_traces", title="HIP API", color="MediumSeaGreen", data=df_hip))
+    result["df_gpu"] = df_gpu
+    result["df_rccl"] = df_rccl
+    result["df_hip"] = df_hip
+    result["sgt_files"] = sgt_files
+
+    # ---------------- packets (optional sniffer output) ----------------
+    try:
+        from . import net as net_mod
+
+        df_pkt = net_mod.parse_pktcap(logdir, tb, cfg)
+        if len(df_pkt):
+            write_trace_csv(df_pkt, os.path.join(logdir, "nettrace.csv"))
+            traces.append(SOFATrace(name="net_traces", title="Network packets", color="OliveDrab", data=df_pkt))
+            traces += _filter_traces(df_pkt, cfg.net_filters, "net")
+        result["df_net"] = df_pkt
+    except Exception as e:
+        p.print_warning(f"packet parse failed: {e}")
+        result["df_net"] = new_trace_df(0)
+
+    # ---------------- report.js ----------------
+    traces_to_json(traces, os.path.join(logdir, "report.js"), plot_ratio=cfg.plot_ratio)
+    n_events = sum(len(t.data) for t in traces if t.data is not None)
+    p.print_progress(f"preprocess done: {len(traces)} series, {n_events} points")
+    result["traces"] = traces
+    return result
