@@ -1,0 +1,73 @@
+"""End-to-end CLI tests (CPU-only host; mirrors the reference's e2e matrix
+test/test.py:62-78: run the real CLI, assert the Complete!! sentinel)."""
+
+import os
+import subprocess
+import sys
+
+import pytest
+
+SOFA = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "bin", "sofa")
+
+
+def run_sofa(args, timeout=120):
+    return subprocess.run(
+        [sys.executable, SOFA] + args,
+        capture_output=True,
+        text=True,
+        timeout=timeout,
+    )
+
+
+def test_stat_dd_complete(tmp_path, native_built):
+    """BASELINE config 1: sofa stat "dd ..." CPU-only plumbing."""
+    logdir = str(tmp_path / "sofalog")
+    out_file = str(tmp_path / "dummy.out")
+    r = run_sofa(
+        [
+            "stat",
+            f"dd if=/dev/zero of={out_file} bs=10M count=20",
+            "--logdir",
+            logdir,
+            "--no_gpu",
+        ]
+    )
+    assert r.returncode == 0, r.stderr
+    assert "Complete!!" in r.stdout
+    for artifact in [
+        "cputrace.csv",
+        "mpstat.csv",
+        "report.js",
+        "features.csv",
+        "performance.csv",
+        "misc.txt",
+    ]:
+        assert os.path.isfile(os.path.join(logdir, artifact)), artifact
+    # sofaboard copied into logdir
+    assert os.path.isfile(os.path.join(logdir, "index.html"))
+
+
+def test_report_rerunnable(tmp_path, native_built):
+    """Stages are idempotent over the logdir (reference --skip_preprocess)."""
+    logdir = str(tmp_path / "sofalog")
+    r = run_sofa(["stat", "sleep 1", "--logdir", logdir, "--no_gpu"])
+    assert "Complete!!" in r.stdout, r.stderr
+    r2 = run_sofa(["report", "--logdir", logdir, "--no_gpu"])
+    assert "Complete!!" in r2.stdout, r2.stderr
+    r3 = run_sofa(["analyze", "--logdir", logdir, "--no_gpu", "--skip_preprocess"])
+    assert "Complete!!" in r3.stdout, r3.stderr
+
+
+def test_clean(tmp_path, native_built):
+    logdir = str(tmp_path / "sofalog")
+    r = run_sofa(["stat", "sleep 0.5", "--logdir", logdir, "--no_gpu"])
+    assert "Complete!!" in r.stdout
+    r = run_sofa(["clean", "--logdir", logdir])
+    assert r.returncode == 0
+    assert not os.path.isfile(os.path.join(logdir, "cputrace.csv"))
+    assert not os.path.isfile(os.path.join(logdir, "report.js"))
+
+
+def test_record_needs_command():
+    r = run_sofa(["record"])
+    assert r.returncode == 2
