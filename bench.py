@@ -1,0 +1,218 @@
+#!/usr/bin/env python3
+"""bench.py — the BASELINE.json headline metric on MI355X:
+
+    profiling overhead (%) + trace events/sec, ResNet-50 DDP at 1/2/4/8 GPUs
+
+Methodology mirrors the reference's overhead-validation harness
+(cyliustack/sofa validation/framework_eval.py:50-99,209-215): the SAME
+workload is timed with and without the profiler; overhead% =
+100 * (t_profiled - t_plain) / t_plain.  Here both phases run in ONE process
+(the collector library supports deferred start + runtime start/stop), so the
+comparison is free of process-restart noise.
+
+Workload: ResNet-50 (own impl, random init) bs=64/GPU, synthetic ImageNet-
+shaped data, bf16 autocast, SGD, DDP over RCCL when WORLD_SIZE > 1.
+
+The JSON `value` is the whole-job trace events/sec while profiling
+(higher is better); `overhead_pct` and plain/profiled ms/step are in
+`config`.  `ms_per_step` is the PROFILED step time (the measured workload).
+
+Run:  python bench.py --gpus N --steps K --warmup W
+DDP:  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+          --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+
+import argparse
+import ctypes
+import glob
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+TRACER = os.path.join(REPO, "sofa_amd", "native", "lib", "libsofatracer.so")
+
+
+def setup_tracer_env(logdir: str) -> None:
+    """Must run BEFORE importing torch (the HIP runtime registers tools at
+    init)."""
+    os.environ["SOFA_LOGDIR"] = logdir
+    os.environ["SOFA_DEFER_START"] = "1"
+    os.environ["SOFA_TRACE_HIP_API"] = "1"
+    os.environ["SOFA_TRACE_RCCL"] = "1"
+    os.environ.setdefault("SOFA_GPU_BUFFER_MB", "64")
+    prev = os.environ.get("ROCP_TOOL_LIBRARIES", "")
+    if TRACER not in prev:
+        os.environ["ROCP_TOOL_LIBRARIES"] = TRACER + ((":" + prev) if prev else "")
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=64)
+    ap.add_argument("--no-profile", action="store_true", help="skip the profiled phase")
+    args = ap.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    logdir = os.path.join(REPO, "gpurun_out", "bench_sgt")
+    if rank == 0:
+        os.makedirs(logdir, exist_ok=True)
+    have_tracer = os.path.exists(TRACER)
+    if have_tracer:
+        setup_tracer_env(logdir)
+
+    import torch  # AFTER env setup
+    import torch.distributed as dist
+    import torch.nn as nn
+
+    use_cuda = torch.cuda.is_available()
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+    distributed = world_size > 1
+    if distributed:
+        dist.init_process_group(
+            backend="nccl" if use_cuda else "gloo",
+            init_method="env://",
+        )
+
+    from sofa_amd.workloads.resnet import build_resnet50
+
+    torch.manual_seed(1234 + rank)
+    model = build_resnet50(device=device, channels_last=use_cuda)
+    if distributed:
+        model = torch.nn.parallel.DistributedDataParallel(
+            model, device_ids=[local_rank] if use_cuda else None,
+            bucket_cap_mb=64,  # fewer, larger RCCL buckets for per-link xGMI efficiency
+        )
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    scaler = None
+    bs = args.batch
+    x = torch.randn(bs, 3, 224, 224, device=device)
+    if use_cuda:
+        x = x.to(memory_format=torch.channels_last)
+    target = torch.randint(0, 1000, (bs,), device=device)
+    loss_fn = nn.CrossEntropyLoss()
+
+    amp_dtype = torch.bfloat16
+    amp_device = "cuda" if use_cuda else "cpu"
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        with torch.autocast(device_type=amp_device, dtype=amp_dtype):
+            loss = loss_fn(model(x), target)
+        loss.backward()
+        opt.step()
+        return loss
+
+    def barrier_sync():
+        if distributed:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    def timed_phase(k):
+        barrier_sync()
+        t0 = time.perf_counter()
+        for _ in range(k):
+            step()
+        barrier_sync()
+        t1 = time.perf_counter()
+        return t1 - t0
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+
+    # ---- phase A: plain (no profiler) ----
+    t_plain = timed_phase(args.steps)
+
+    # ---- phase B: profiled (collector armed + SysMonitor + cpusampler) ----
+    t_prof = None
+    n_events = 0
+    lib = None
+    if have_tracer and use_cuda and not args.no_profile:
+        lib = ctypes.CDLL(TRACER)
+        lib.sofa_tracer_event_count.restype = ctypes.c_ulonglong
+        from sofa_amd.record.pollers import SysMonitor
+        import subprocess
+
+        mon = SysMonitor(logdir, rate_hz=10, enable_gpu=(local_rank == 0))
+        mon.start()
+        sampler_bin = os.path.join(REPO, "sofa_amd", "native", "bin", "sofa-cpusampler")
+        sampler = None
+        if os.path.exists(sampler_bin):
+            sampler = subprocess.Popen(
+                [sampler_bin, "-o", os.path.join(logdir, f"bench_{rank}.scs"),
+                 "-F", "99", "-p", str(os.getpid())]
+            )
+        n0 = lib.sofa_tracer_event_count()
+        lib.sofa_tracer_start()
+        t_prof = timed_phase(args.steps)
+        lib.sofa_tracer_stop()
+        n_events = int(lib.sofa_tracer_event_count() - n0)
+        if sampler is not None:
+            sampler.terminate()
+            sampler.wait(timeout=5)
+        mon.stop()
+        mon.join(timeout=5)
+    else:
+        # CPU smoke path: rerun plain so the output shape stays identical
+        t_prof = timed_phase(args.steps)
+
+    # ---- aggregate across ranks: MAX time, SUM events ----
+    if distributed:
+        tt = torch.tensor([t_plain, t_prof], dtype=torch.float64, device=device if use_cuda else None)
+        dist.all_reduce(tt, op=dist.ReduceOp.MAX)
+        t_plain, t_prof = tt[0].item(), tt[1].item()
+        ev = torch.tensor([float(n_events)], dtype=torch.float64, device=device if use_cuda else None)
+        dist.all_reduce(ev, op=dist.ReduceOp.SUM)
+        n_events = int(ev[0].item())
+
+    overhead_pct = 100.0 * (t_prof - t_plain) / t_plain if t_plain > 0 else 0.0
+    events_per_sec = n_events / t_prof if t_prof > 0 else 0.0
+    imgs_per_sec = world_size * bs * args.steps / t_prof if t_prof > 0 else 0.0
+
+    if rank == 0:
+        result = {
+            "metric": "trace_events_per_sec (ResNet-50 DDP under full profiling)",
+            "value": round(events_per_sec, 1),
+            "unit": "events/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(t_prof / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "resnet50",
+                "global_batch": world_size * bs,
+                "seq_len": None,
+                "parallelism": f"dp{world_size}",
+                "image_size": 224,
+                "profiling_overhead_pct": round(overhead_pct, 3),
+                "ms_per_step_plain": round(t_plain / args.steps * 1e3, 3),
+                "images_per_sec_profiled": round(imgs_per_sec, 1),
+                "trace_events": n_events,
+            },
+        }
+        print(json.dumps(result))
+    if distributed:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

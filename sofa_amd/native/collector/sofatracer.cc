@@ -489,7 +489,9 @@ int tool_init(rocprofiler_client_finalize_t, void*) {
   int valid = 0;
   rocprofiler_context_is_valid(g_ctx, &valid);
   if (valid == 0) return -1;
-  rocprofiler_start_context(g_ctx);
+  // SOFA_DEFER_START=1: stay disarmed until sofa_tracer_start() (bench.py's
+  // in-process overhead A/B measurement)
+  if (!env_flag("SOFA_DEFER_START", false)) rocprofiler_start_context(g_ctx);
   return 0;
 }
 
@@ -504,6 +506,27 @@ void tool_fini(void*) {
 }
 
 }  // namespace
+
+// ---- runtime control API (dlopen'd by bench.py / sofa_amd via ctypes) ----
+
+extern "C" int sofa_tracer_start() {
+  if (g_ctx.handle == 0) return -1;
+  return rocprofiler_start_context(g_ctx) == ROCPROFILER_STATUS_SUCCESS ? 0 : -1;
+}
+
+extern "C" int sofa_tracer_stop() {
+  if (g_ctx.handle == 0) return -1;
+  auto st = rocprofiler_stop_context(g_ctx);
+  rocprofiler_flush_buffer(g_buffer);
+  return st == ROCPROFILER_STATUS_SUCCESS ? 0 : -1;
+}
+
+extern "C" unsigned long long sofa_tracer_event_count() {
+  rocprofiler_flush_buffer(g_buffer);
+  return g_n_records.load(std::memory_order_relaxed);
+}
+
+extern "C" int sofa_tracer_active() { return g_out != nullptr; }
 
 extern "C" rocprofiler_tool_configure_result_t*
 rocprofiler_configure(uint32_t version, const char* runtime_version,
