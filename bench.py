@@ -56,6 +56,9 @@ def main() -> int:
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--no-profile", action="store_true", help="skip the profiled phase")
+    ap.add_argument("--hip-api", type=int, default=1, help="trace HIP runtime API (0/1)")
+    ap.add_argument("--full-record", type=int, default=1,
+                    help="also run SysMonitor + cpusampler during profiled phase")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -68,6 +71,7 @@ def main() -> int:
     have_tracer = os.path.exists(TRACER)
     if have_tracer:
         setup_tracer_env(logdir)
+        os.environ["SOFA_TRACE_HIP_API"] = "1" if args.hip_api else "0"
 
     import torch  # AFTER env setup
     import torch.distributed as dist
@@ -146,25 +150,33 @@ def main() -> int:
         from sofa_amd.record.pollers import SysMonitor
         import subprocess
 
-        mon = SysMonitor(logdir, rate_hz=10, enable_gpu=(local_rank == 0))
-        mon.start()
-        sampler_bin = os.path.join(REPO, "sofa_amd", "native", "bin", "sofa-cpusampler")
+        mon = None
         sampler = None
-        if os.path.exists(sampler_bin):
-            sampler = subprocess.Popen(
-                [sampler_bin, "-o", os.path.join(logdir, f"bench_{rank}.scs"),
-                 "-F", "99", "-p", str(os.getpid())]
-            )
-        n0 = lib.sofa_tracer_event_count()
+        if args.full_record:
+            mon = SysMonitor(logdir, rate_hz=10, enable_gpu=(local_rank == 0))
+            mon.start()
+            sampler_bin = os.path.join(REPO, "sofa_amd", "native", "bin", "sofa-cpusampler")
+            if os.path.exists(sampler_bin):
+                sampler = subprocess.Popen(
+                    [sampler_bin, "-o", os.path.join(logdir, f"bench_{rank}.scs"),
+                     "-F", "99", "-p", str(os.getpid())]
+                )
         lib.sofa_tracer_start()
+        # profiled warmup: first traced launches pay one-time interception
+        # setup; keep that out of the timed region
+        for _ in range(max(2, args.warmup // 2)):
+            step()
+        barrier_sync()
+        n0 = lib.sofa_tracer_event_count()
         t_prof = timed_phase(args.steps)
         lib.sofa_tracer_stop()
         n_events = int(lib.sofa_tracer_event_count() - n0)
         if sampler is not None:
             sampler.terminate()
             sampler.wait(timeout=5)
-        mon.stop()
-        mon.join(timeout=5)
+        if mon is not None:
+            mon.stop()
+            mon.join(timeout=5)
     else:
         # CPU smoke path: rerun plain so the output shape stays identical
         t_prof = timed_phase(args.steps)
