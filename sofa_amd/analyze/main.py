@@ -136,18 +136,26 @@ def sofa_analyze(cfg: SofaConfig, pre: Optional[dict] = None) -> Dict[str, float
     print("\nPerformance features:")
     print(fdf.to_string(index=False))
 
-    # --- advisor (POTATO-parity gRPC, reference :1007-1048) ---
-    if cfg.potato_server:
-        try:
-            from ..advisor.client import get_hint
+    # --- advisor (POTATO-parity gRPC + built-in rule engine,
+    #     reference :1007-1048) ---
+    try:
+        from ..advisor.client import get_hint, local_hints
 
-            hint = get_hint(cfg.potato_server, fdf)
-            if hint:
-                p.print_hint(hint)
-                with open(os.path.join(logdir, "potato_report.html"), "w") as f:
-                    f.write("<html><body><pre>%s</pre></body></html>" % hint)
-        except Exception as e:
-            p.print_warning(f"advisor unavailable: {e}")
+        hint = None
+        if cfg.potato_server:
+            try:
+                hint = get_hint(cfg.potato_server, fdf)
+            except Exception as e:
+                p.print_warning(f"advisor server unavailable ({e}); using local rules")
+        if hint is None:
+            hint = local_hints(fdf)
+        if hint:
+            print("\nAdvisor hints:")
+            p.print_hint(hint)
+            with open(os.path.join(logdir, "potato_report.html"), "w") as f:
+                f.write("<html><body><pre>%s</pre></body></html>" % hint)
+    except Exception as e:
+        p.print_warning(f"advisor failed: {e}")
 
     # --- copy sofaboard into logdir (reference :1050-1052) ---
     board_src = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "sofaboard")
