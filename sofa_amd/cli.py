@@ -50,6 +50,8 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--sys_mon_rate", type=int, default=10)
     ap.add_argument("--profile_all_cpus", action="store_true")
     ap.add_argument("--enable_tcpdump", action="store_true")
+    ap.add_argument("--enable_strace", action="store_true")
+    ap.add_argument("--enable_py_stacks", action="store_true")
     ap.add_argument("--no_gpu", action="store_true", help="disable GPU tracing")
     ap.add_argument("--no_hip_api", action="store_true")
     ap.add_argument("--no_rccl", action="store_true")
@@ -91,6 +93,8 @@ def cfg_from_args(args) -> SofaConfig:
         sys_mon_rate=args.sys_mon_rate,
         profile_all_cpus=args.profile_all_cpus,
         enable_tcpdump=args.enable_tcpdump,
+        enable_strace=args.enable_strace,
+        enable_pystacks=args.enable_py_stacks,
         enable_gpu=not args.no_gpu,
         enable_gpu_hip_api=not args.no_hip_api,
         enable_rccl_trace=not args.no_rccl,

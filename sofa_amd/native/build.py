@@ -55,6 +55,12 @@ def build_all(verbose: bool = False) -> None:
             [],
         ),
         (
+            os.path.join(BIN, "sofa-syscalltrace"),
+            [os.path.join(HERE, "syscalltrace", "syscalltrace.cc")],
+            ["g++", "-O2", "-std=c++17"],
+            [],
+        ),
+        (
             os.path.join(LIB, "libsofatracer.so"),
             [
                 os.path.join(HERE, "collector", "sofatracer.cc"),

@@ -178,6 +178,32 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
         p.print_warning(f"packet parse failed: {e}")
         result["df_net"] = new_trace_df(0)
 
+    # ---------------- syscalls (SST from sofa-syscalltrace) ----------------
+    try:
+        from .strace import parse_sst
+
+        df_strace = parse_sst(logdir, tb, cfg)
+        if len(df_strace):
+            write_trace_csv(df_strace, os.path.join(logdir, "strace.csv"))
+            traces.append(SOFATrace(name="strace_traces", title="Syscalls", color="DarkKhaki", data=df_strace))
+        result["df_strace"] = df_strace
+    except Exception as e:
+        p.print_warning(f"syscall trace parse failed: {e}")
+        result["df_strace"] = new_trace_df(0)
+
+    # ---------------- python stacks ----------------
+    try:
+        from .pystacks import parse_pystacks
+
+        df_py = parse_pystacks(logdir, tb)
+        if len(df_py):
+            write_trace_csv(df_py, os.path.join(logdir, "pystacks.csv"))
+            traces.append(SOFATrace(name="pystacks_traces", title="Python stacks", color="SteelBlue", data=df_py))
+        result["df_pystacks"] = df_py
+    except Exception as e:
+        p.print_warning(f"pystacks parse failed: {e}")
+        result["df_pystacks"] = new_trace_df(0)
+
     # ---------------- report.js ----------------
     traces_to_json(traces, os.path.join(logdir, "report.js"), plot_ratio=cfg.plot_ratio)
     n_events = sum(len(t.data) for t in traces if t.data is not None)

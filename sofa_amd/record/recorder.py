@@ -201,9 +201,23 @@ def sofa_record(command: str, cfg: SofaConfig) -> int:
     # --- launch target ---
     env = build_target_env(cfg)
     err_f = open(os.path.join(logdir, "sofa.err"), "w")
+    argv = ["bash", "-c", command]
+    if cfg.enable_strace:
+        st_bin = native_bin("sofa-syscalltrace")
+        if os.path.exists(st_bin):
+            argv = [st_bin, "-o", os.path.join(logdir, "strace.sst"), "--"] + argv
+        else:
+            p.print_warning("sofa-syscalltrace not built; syscall tracing disabled")
+    if cfg.enable_pystacks:
+        # inject the in-process Python stack sampler via PYTHONPATH
+        # (pyflame-replacement; see sofa_amd/pystacks_inject)
+        inject_dir = os.path.join(os.path.dirname(native_dir()), "pystacks_inject")
+        if os.path.isdir(inject_dir):
+            env["PYTHONPATH"] = inject_dir + ":" + env.get("PYTHONPATH", "")
+            env["SOFA_PYSTACKS_OUT"] = os.path.join(logdir, "pystacks.txt")
     t_begin = time.time()
     target = subprocess.Popen(
-        ["bash", "-c", command],
+        argv,
         env=env,
         stdout=None,
         stderr=err_f if not cfg.verbose else None,
