@@ -56,7 +56,9 @@ def main() -> int:
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--no-profile", action="store_true", help="skip the profiled phase")
-    ap.add_argument("--hip-api", type=int, default=1, help="trace HIP runtime API (0/1)")
+    ap.add_argument("--hip-api", type=int, default=0,
+                    help="trace HIP runtime API (0/1; off by default, matching "
+                    "the reference's opt-in --cuda_api_tracing)")
     ap.add_argument("--full-record", type=int, default=1,
                     help="also run SysMonitor + cpusampler during profiled phase")
     args = ap.parse_args()
@@ -147,14 +149,19 @@ def main() -> int:
     if have_tracer and use_cuda and not args.no_profile:
         lib = ctypes.CDLL(TRACER)
         lib.sofa_tracer_event_count.restype = ctypes.c_ulonglong
-        from sofa_amd.record.pollers import SysMonitor
         import subprocess
 
         mon = None
         sampler = None
         if args.full_record:
-            mon = SysMonitor(logdir, rate_hz=10, enable_gpu=(local_rank == 0))
-            mon.start()
+            # telemetry monitor as a SUBPROCESS (the real `sofa record`
+            # architecture: pollers live in the recorder, not the target)
+            if local_rank == 0:
+                mon = subprocess.Popen(
+                    [sys.executable, "-m", "sofa_amd.record.monitor_main",
+                     "--logdir", logdir, "--rate", "10", "--parent", str(os.getpid())],
+                    cwd=REPO,
+                )
             sampler_bin = os.path.join(REPO, "sofa_amd", "native", "bin", "sofa-cpusampler")
             if os.path.exists(sampler_bin):
                 sampler = subprocess.Popen(
@@ -175,8 +182,8 @@ def main() -> int:
             sampler.terminate()
             sampler.wait(timeout=5)
         if mon is not None:
-            mon.stop()
-            mon.join(timeout=5)
+            mon.terminate()
+            mon.wait(timeout=5)
     else:
         # CPU smoke path: rerun plain so the output shape stays identical
         t_prof = timed_phase(args.steps)

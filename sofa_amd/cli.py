@@ -53,7 +53,12 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--enable_strace", action="store_true")
     ap.add_argument("--enable_py_stacks", action="store_true")
     ap.add_argument("--no_gpu", action="store_true", help="disable GPU tracing")
-    ap.add_argument("--no_hip_api", action="store_true")
+    ap.add_argument(
+        "--hip_api_trace",
+        action="store_true",
+        help="also trace HIP runtime API spans (opt-in; adds ~45%% overhead "
+        "on launch-dense workloads)",
+    )
     ap.add_argument("--no_rccl", action="store_true")
     ap.add_argument("--gpu_buffer_mb", type=int, default=64)
     # preprocess
@@ -96,7 +101,7 @@ def cfg_from_args(args) -> SofaConfig:
         enable_strace=args.enable_strace,
         enable_pystacks=args.enable_py_stacks,
         enable_gpu=not args.no_gpu,
-        enable_gpu_hip_api=not args.no_hip_api,
+        enable_gpu_hip_api=args.hip_api_trace,
         enable_rccl_trace=not args.no_rccl,
         gpu_ring_buffer_mb=args.gpu_buffer_mb,
         cpu_time_offset_ms=args.cpu_time_offset_ms,

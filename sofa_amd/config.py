@@ -79,7 +79,10 @@ class SofaConfig:
     enable_diskstat: bool = True
     enable_netstat: bool = True
     enable_gpu: bool = True            # rocprofiler-sdk collector
-    enable_gpu_hip_api: bool = True    # also record HIP runtime API spans
+    # HIP runtime API spans are OPT-IN (reference parity: --cuda_api_tracing
+    # was opt-in too, bin/sofa:*) — measured +~45% overhead on launch-dense
+    # training steps (profiles/overhead_matrix notes)
+    enable_gpu_hip_api: bool = False
     enable_rccl_trace: bool = True     # RCCL API tracing via collector
     enable_kfd_trace: bool = False     # page-migrate/fault events
     gpu_ring_buffer_mb: int = 64       # collector buffer size per process

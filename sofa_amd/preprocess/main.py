@@ -156,7 +156,7 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
     if len(df_rccl):
         write_trace_csv(df_rccl, os.path.join(logdir, "rccltrace.csv"))
         traces.append(SOFATrace(name="rccl_traces", title="RCCL collectives", color="Crimson", data=df_rccl))
-    if len(df_hip) and cfg.enable_gpu_hip_api:
+    if len(df_hip):
         write_trace_csv(df_hip, os.path.join(logdir, "hip_api_trace.csv"))
         traces.append(SOFATrace(name="hip_api_traces", title="HIP API", color="MediumSeaGreen", data=df_hip))
     result["df_gpu"] = df_gpu
