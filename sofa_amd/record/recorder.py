@@ -44,6 +44,8 @@ RAW_FILES = [
     "kallsyms",
     "misc.txt",
     "pktcap.bin",
+    "strace.sst",
+    "sofa_pids.txt",
     "sofa.err",
 ]
 
@@ -110,8 +112,8 @@ def sofa_clean(cfg: SofaConfig) -> None:
         if (
             name in RAW_FILES
             or name in DERIVED_FILES
-            or name.startswith(("gputrace_", "sofa_hints"))
-            or name.endswith((".sgt", ".scs"))
+            or name.startswith(("gputrace_", "sofa_hints", "pystacks.txt"))
+            or name.endswith((".sgt", ".scs", ".sst"))
         ):
             if os.path.isdir(full):
                 shutil.rmtree(full, ignore_errors=True)
@@ -239,6 +241,14 @@ def sofa_record(command: str, cfg: SofaConfig) -> int:
             p.print_warning(f"cpusampler failed to start: {e}")
     else:
         p.print_warning("sofa-cpusampler not built; CPU sampling disabled")
+
+    # pid inventory for tools/killsofa.sh (kill exact PIDs, never patterns)
+    with open(os.path.join(logdir, "sofa_pids.txt"), "w") as f:
+        f.write("%d\n" % target.pid)
+        if sampler is not None:
+            f.write("%d\n" % sampler.pid)
+        if pktcap_proc is not None:
+            f.write("%d\n" % pktcap_proc.pid)
 
     # --- wait ---
     try:
