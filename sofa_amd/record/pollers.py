@@ -139,8 +139,7 @@ class SysMonitor(threading.Thread):
                 self.smi = None
         use_gpu = self.smi is not None and self.smi.available
 
-        next_t = time.time()
-        while not self.stop_event.is_set():
+        def tick():
             ts = time.time()
             mhz = read_cpuinfo_mhz()
             if mhz:
@@ -184,12 +183,19 @@ class SysMonitor(threading.Thread):
                             power if power is not None else -1.0,
                         )
                     )
+
+        next_t = time.time()
+        while not self.stop_event.is_set():
+            tick()
             next_t += self.period
             delay = next_t - time.time()
             if delay > 0:
                 self.stop_event.wait(delay)
             else:
                 next_t = time.time()
+        # one final sample so even sub-period runs get >=2 ticks (delta
+        # parsers need two snapshots)
+        tick()
 
         for f in self._files.values():
             try:
