@@ -86,7 +86,24 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--with-gui", dest="with_gui", action="store_true")
     # plugins (reference bin/sofa:21,322)
     ap.add_argument("--plugins", default="", help="comma-separated module names; each must expose f(cfg)")
+    # declarative config file (successor of the reference's vestigial
+    # examples/conf/*.cfg libconfig surface)
+    ap.add_argument("--config", default="", help="YAML file of SofaConfig fields (CLI flags win)")
     return ap
+
+
+def apply_config_file(cfg: SofaConfig, path: str) -> None:
+    import yaml
+
+    with open(path) as f:
+        data = yaml.safe_load(f) or {}
+    for key, val in data.items():
+        if key in ("cpu_filters", "gpu_filters", "net_filters", "diskstat_filters"):
+            val = [Filter(**v) if isinstance(v, dict) else Filter(str(v), "") for v in val]
+        if hasattr(cfg, key):
+            setattr(cfg, key, val)
+        else:
+            p.print_warning(f"config: unknown key {key}")
 
 
 def cfg_from_args(args) -> SofaConfig:
@@ -145,6 +162,12 @@ def run_plugins(args, cfg) -> None:
 def main(argv=None) -> int:
     args = build_parser().parse_args(argv)
     cfg = cfg_from_args(args)
+    if args.config:
+        try:
+            apply_config_file(cfg, args.config)
+        except Exception as e:
+            p.print_error(f"bad --config {args.config}: {e}")
+            return 2
     run_plugins(args, cfg)
     verb = args.command_verb
 
