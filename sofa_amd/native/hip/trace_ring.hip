@@ -227,4 +227,47 @@ int sofa_ring_compact(void* ring_p, uint64_t tag_mask, double scale,
   return 0;
 }
 
+// Kernel-only throughput benchmark: run the compaction kernel `iters` times
+// (device buffers only, no D2H), return mean kernel ms via hipEvents.
+int sofa_ring_compact_bench(void* ring_p, uint64_t tag_mask, int iters,
+                            double* ms_out) {
+  Ring* ring = static_cast<Ring*>(ring_p);
+  HIP_CHECK(hipSetDevice(ring->device));
+  RingControl ctl;
+  HIP_CHECK(hipMemcpy(&ctl, ring->d_ctl, sizeof(ctl), hipMemcpyDeviceToHost));
+  uint32_t n_valid =
+      (uint32_t)(ctl.head < ctl.capacity ? ctl.head : ctl.capacity);
+  if (n_valid == 0) return -1;
+  RingRec* d_out = nullptr;
+  unsigned int* d_count = nullptr;
+  HIP_CHECK(hipMalloc(&d_out, sizeof(RingRec) * (size_t) n_valid));
+  HIP_CHECK(hipMalloc(&d_count, sizeof(unsigned int)));
+  dim3 block(kBlockThreads);
+  dim3 grid((n_valid + kBlockThreads - 1) / kBlockThreads);
+  // warmup
+  HIP_CHECK(hipMemset(d_count, 0, sizeof(unsigned int)));
+  hipLaunchKernelGGL(compact_kernel, grid, block, 0, 0, ring->d_slots, n_valid,
+                     tag_mask, 10.0, 12345, d_out, n_valid, d_count);
+  HIP_CHECK(hipDeviceSynchronize());
+  hipEvent_t e0, e1;
+  HIP_CHECK(hipEventCreate(&e0));
+  HIP_CHECK(hipEventCreate(&e1));
+  HIP_CHECK(hipEventRecord(e0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(compact_kernel, grid, block, 0, 0, ring->d_slots,
+                       n_valid, tag_mask, 10.0, 12345, d_out, n_valid,
+                       d_count);
+  }
+  HIP_CHECK(hipEventRecord(e1));
+  HIP_CHECK(hipDeviceSynchronize());
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+  *ms_out = ms / iters;
+  HIP_CHECK(hipFree(d_out));
+  HIP_CHECK(hipFree(d_count));
+  HIP_CHECK(hipEventDestroy(e0));
+  HIP_CHECK(hipEventDestroy(e1));
+  return 0;
+}
+
 }  // extern "C"
