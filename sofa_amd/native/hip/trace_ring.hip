@@ -84,8 +84,13 @@ __global__ void producer_kernel(RingControl* ctl, RingRec* slots, uint32_t n,
 // scale is ~10.0 for the 100 MHz counter, exact for the < 2^53 tick values
 // seen in practice).
 
-constexpr int kBlockThreads = 256;          // 4 waves
+constexpr int kBlockThreads = 256;          // 4 waves of 64
 constexpr int kWavesPerBlock = kBlockThreads / 64;
+constexpr int kRecsPerThread = 8;           // 2048 records per block: the
+// original 1-record/thread version spent its time on the 2 __syncthreads +
+// 1 global atomic per 256 records (measured 0.9 TB/s); amortizing them over
+// 8x the data and scanning per-lane COUNTS (0..8) instead of single-bit
+// ballots keeps the output stable in input order.
 
 __global__ void compact_kernel(const RingRec* __restrict__ slots, uint32_t n_valid,
                                uint64_t tag_mask, double scale, long long offset,
@@ -95,24 +100,37 @@ __global__ void compact_kernel(const RingRec* __restrict__ slots, uint32_t n_val
   __shared__ uint32_t wave_bases[kWavesPerBlock];
   __shared__ unsigned int block_base;
 
-  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
+  // thread t owns records [base + t*R, base + t*R + R): consecutive per
+  // thread so the kept-record order equals input order after the scan.
+  uint32_t first = (blockIdx.x * kBlockThreads + threadIdx.x) * kRecsPerThread;
 
-  RingRec r{};
-  bool keep = false;
-  if (i < n_valid) {
-    r = slots[i];
-    keep = r.tag != 0 && ((tag_mask >> (r.tag & 63u)) & 1ull);
+  RingRec r[kRecsPerThread];
+  bool keep[kRecsPerThread];
+  uint32_t my_count = 0;
+#pragma unroll
+  for (int j = 0; j < kRecsPerThread; ++j) {
+    uint32_t i = first + j;
+    keep[j] = false;
+    if (i < n_valid) {
+      r[j] = slots[i];
+      keep[j] = r[j].tag != 0 && ((tag_mask >> (r[j].tag & 63u)) & 1ull);
+    }
+    my_count += keep[j] ? 1u : 0u;
   }
 
-  // 64-wide ballot compaction (CDNA: __ballot returns a 64-bit mask);
-  // ((1ull<<lane)-1) is well-defined for lane in [0,63]
-  unsigned long long mask = __ballot(keep);
-  uint32_t prefix = __popcll(mask & ((1ull << lane) - 1ull));
-  uint32_t wave_total = __popcll(mask);
+  // wave-level exclusive scan of per-lane counts (log2(64) shuffle steps)
+  uint32_t scan = my_count;
+#pragma unroll
+  for (int d = 1; d < 64; d <<= 1) {
+    uint32_t up = __shfl_up(scan, d, 64);
+    if (lane >= d) scan += up;
+  }
+  uint32_t lane_base = scan - my_count;               // exclusive
+  uint32_t wave_total = __shfl(scan, 63, 64);         // inclusive of lane 63
 
-  if (lane == 0) wave_totals[wave] = wave_total;
+  if (lane == 63) wave_totals[wave] = wave_total;
   __syncthreads();
 
   if (wave == 0 && lane == 0) {
@@ -125,12 +143,15 @@ __global__ void compact_kernel(const RingRec* __restrict__ slots, uint32_t n_val
   }
   __syncthreads();
 
-  if (keep) {
-    uint32_t dst = block_base + wave_bases[wave] + prefix;
-    if (dst < out_cap) {
-      r.t_start = (uint64_t)((double) r.t_start * scale + (double) offset);
-      r.t_end = (uint64_t)((double) r.t_end * scale + (double) offset);
-      out[dst] = r;
+  uint32_t dst = block_base + wave_bases[wave] + lane_base;
+#pragma unroll
+  for (int j = 0; j < kRecsPerThread; ++j) {
+    if (keep[j] && dst < out_cap) {
+      RingRec o = r[j];
+      o.t_start = (uint64_t)((double) o.t_start * scale + (double) offset);
+      o.t_end = (uint64_t)((double) o.t_end * scale + (double) offset);
+      out[dst] = o;
+      ++dst;
     }
   }
 }
@@ -207,7 +228,8 @@ int sofa_ring_compact(void* ring_p, uint64_t tag_mask, double scale,
 
   if (n_valid > 0) {
     dim3 block(kBlockThreads);
-    dim3 grid((n_valid + kBlockThreads - 1) / kBlockThreads);
+    uint32_t per_block = kBlockThreads * kRecsPerThread;
+    dim3 grid((n_valid + per_block - 1) / per_block);
     hipLaunchKernelGGL(compact_kernel, grid, block, 0, 0, ring->d_slots,
                        n_valid, tag_mask, scale, offset, d_out, n_valid,
                        d_count);
@@ -243,7 +265,8 @@ int sofa_ring_compact_bench(void* ring_p, uint64_t tag_mask, int iters,
   HIP_CHECK(hipMalloc(&d_out, sizeof(RingRec) * (size_t) n_valid));
   HIP_CHECK(hipMalloc(&d_count, sizeof(unsigned int)));
   dim3 block(kBlockThreads);
-  dim3 grid((n_valid + kBlockThreads - 1) / kBlockThreads);
+  uint32_t per_block = kBlockThreads * kRecsPerThread;
+  dim3 grid((n_valid + per_block - 1) / per_block);
   // warmup
   HIP_CHECK(hipMemset(d_count, 0, sizeof(unsigned int)));
   hipLaunchKernelGGL(compact_kernel, grid, block, 0, 0, ring->d_slots, n_valid,
