@@ -60,22 +60,23 @@ def scs_to_cputrace(
         if kallsyms and not os.path.isfile(kallsyms):
             kallsyms = ""
         symr = Symbolizer(scs.mmaps, kallsyms)
-        # resolve unique (pid, ip, kernel-flag) triples only
-        kern = (s["flags"] & 1).astype(bool)
-        keys = {}
-        names = [""] * n
-        for i in range(n):
-            key = (int(s["pid"][i]), int(s["ip"][i]), bool(kern[i]))
-            nm = keys.get(key)
-            if nm is None:
-                sym, dso = symr.resolve(key[0], key[1], key[2])
-                comm = scs.comms.get(int(s["tid"][i]), "")
-                nm = f"{sym} @ {dso}"
-                if comm:
-                    nm = f"{nm} [{comm}]"
-                keys[key] = nm
-            names[i] = nm
-        df["name"] = names
+        # resolve UNIQUE (pid, ip, kernel-flag) triples only, then map back
+        # vectorized (a long run has millions of samples but few unique IPs)
+        kern = (s["flags"].astype(np.uint64) & 1)
+        key = (s["pid"].astype(np.uint64) << 49) | (kern << 48) | (
+            s["ip"].astype(np.uint64) & ((1 << 48) - 1)
+        )
+        uniq, inv, first_idx = np.unique(key, return_inverse=True, return_index=True)
+        uniq_names = np.empty(len(uniq), dtype=object)
+        for u in range(len(uniq)):
+            i = int(first_idx[u])
+            sym, dso = symr.resolve(int(s["pid"][i]), int(s["ip"][i]), bool(kern[i]))
+            comm = scs.comms.get(int(s["tid"][i]), "")
+            nm = f"{sym} @ {dso}"
+            if comm:
+                nm = f"{nm} [{comm}]"
+            uniq_names[u] = nm
+        df["name"] = uniq_names[inv]
     else:
-        df["name"] = ["ip_0x%x" % ip for ip in s["ip"]]
+        df["name"] = np.char.add("ip_", s["ip"].astype("U16"))
     return df

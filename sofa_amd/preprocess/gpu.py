@@ -230,22 +230,26 @@ def sgt_to_hip_api_trace(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.Dat
         a = sgt.hip_api
         if not len(a):
             continue
-        df = new_trace_df(len(a))
-        ts = _timeline(tb, sgt, a["start_ns"])
-        dur = (a["end_ns"] - a["start_ns"]).astype(np.float64) * 1e-9
         hip_names = {
             op: nm
             for (kind, op), nm in sgt.opnames.items()
             if nm.startswith("hip")
         }
         op = a["op"]
-        df["timestamp"] = ts
-        df["duration"] = dur
-        df["event"] = op.astype(np.float64)
-        df["pid"] = sgt.pid
-        df["tid"] = a["tid"].astype(np.int64)
-        df["name"] = [hip_names.get(int(o), "hip_api_%d" % o) for o in op]
-        df["category"] = 1
+        uniq, inv = np.unique(op, return_inverse=True)
+        uniq_names = np.array(
+            [hip_names.get(int(o), "hip_api_%d" % o) for o in uniq], dtype=object
+        )
+        df = trace_df_from(
+            len(a),
+            timestamp=_timeline(tb, sgt, a["start_ns"]),
+            duration=(a["end_ns"] - a["start_ns"]).astype(np.float64) * 1e-9,
+            event=op.astype(np.float64),
+            pid=np.full(len(a), sgt.pid, dtype=np.int64),
+            tid=a["tid"].astype(np.int64),
+            name=uniq_names[inv],
+            category=np.ones(len(a), dtype=np.int64),
+        )
         frames.append(df)
     if not frames:
         return new_trace_df(0)
