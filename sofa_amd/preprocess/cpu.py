@@ -66,7 +66,8 @@ def scs_to_cputrace(
         key = (s["pid"].astype(np.uint64) << 49) | (kern << 48) | (
             s["ip"].astype(np.uint64) & ((1 << 48) - 1)
         )
-        uniq, inv, first_idx = np.unique(key, return_inverse=True, return_index=True)
+        # np.unique returns (values, first_indices, inverse) in this order
+        uniq, first_idx, inv = np.unique(key, return_index=True, return_inverse=True)
         uniq_names = np.empty(len(uniq), dtype=object)
         for u in range(len(uniq)):
             i = int(first_idx[u])
