@@ -60,6 +60,12 @@ def build_parser() -> argparse.ArgumentParser:
         "on launch-dense workloads)",
     )
     ap.add_argument("--no_rccl", action="store_true")
+    ap.add_argument(
+        "--rccl_shim",
+        action="store_true",
+        help="trace RCCL via the LD_PRELOAD interposer instead of/next to "
+        "the rocprofiler-sdk path",
+    )
     ap.add_argument("--gpu_buffer_mb", type=int, default=64)
     # preprocess
     ap.add_argument("--cpu_time_offset_ms", type=int, default=0)
@@ -120,6 +126,7 @@ def cfg_from_args(args) -> SofaConfig:
         enable_gpu=not args.no_gpu,
         enable_gpu_hip_api=args.hip_api_trace,
         enable_rccl_trace=not args.no_rccl,
+        rccl_shim=args.rccl_shim,
         gpu_ring_buffer_mb=args.gpu_buffer_mb,
         cpu_time_offset_ms=args.cpu_time_offset_ms,
         plot_ratio=args.plot_ratio,

@@ -84,6 +84,7 @@ class SofaConfig:
     # training steps (profiles/overhead_matrix notes)
     enable_gpu_hip_api: bool = False
     enable_rccl_trace: bool = True     # RCCL API tracing via collector
+    rccl_shim: bool = False            # LD_PRELOAD interposer (fallback path)
     enable_kfd_trace: bool = False     # page-migrate/fault events
     gpu_ring_buffer_mb: int = 64       # collector buffer size per process
     blkdev: str = ""                   # block device for blktrace-like stats

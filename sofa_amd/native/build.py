@@ -61,6 +61,15 @@ def build_all(verbose: bool = False) -> None:
             [],
         ),
         (
+            os.path.join(LIB, "libsofarccl.so"),
+            [
+                os.path.join(HERE, "rccl_shim", "rccl_shim.cc"),
+                os.path.join(HERE, "collector", "sgt_format.h"),
+            ],
+            ["g++", "-O2", "-std=c++17", "-fPIC", "-shared"],
+            ["-ldl", "-lpthread"],
+        ),
+        (
             os.path.join(LIB, "libsofatracer.so"),
             [
                 os.path.join(HERE, "collector", "sofatracer.cc"),

@@ -1,0 +1,168 @@
+// libsofarccl — LD_PRELOAD interposer for RCCL collectives.
+//
+// Fallback/complement to the rocprofiler-sdk RCCL API tracing in
+// libsofatracer (SURVEY.md §2.7 "RCCL interception (LD_PRELOAD shim)"):
+// wraps the nccl* entry points via dlsym(RTLD_NEXT), stamps
+// CLOCK_MONOTONIC_RAW enter/exit, and writes RcclRec records in the same SGT
+// container (sgt_format.h) to $SOFA_LOGDIR/rcclshim_<pid>.sgt so preprocess
+// merges them identically.  Zero dependency on rocprofiler-register — works
+// against any librccl.
+//
+// Enable:  LD_PRELOAD=/path/libsofarccl.so SOFA_LOGDIR=<logdir> <cmd>
+
+#include <dlfcn.h>
+#include <pthread.h>
+#include <unistd.h>
+
+#include <cstdint>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <ctime>
+
+#include "../collector/sgt_format.h"
+
+namespace {
+
+using ncclResult_t = int;
+using ncclComm_t = void*;
+using hipStream_t = void*;
+using ncclDataType_t = int;
+using ncclRedOp_t = int;
+
+FILE* g_out = nullptr;
+pthread_mutex_t g_mutex = PTHREAD_MUTEX_INITIALIZER;
+
+uint64_t now_ns(clockid_t c) {
+  struct timespec ts;
+  clock_gettime(c, &ts);
+  return uint64_t(ts.tv_sec) * 1000000000ull + ts.tv_nsec;
+}
+
+uint32_t elem_size(int dt) {
+  static const uint32_t sz[] = {1, 1, 4, 4, 8, 8, 2, 4, 8, 2, 1, 1};
+  return dt >= 0 && dt < (int) (sizeof(sz) / sizeof(sz[0])) ? sz[dt] : 0;
+}
+
+void ensure_open() {
+  if (g_out) return;
+  pthread_mutex_lock(&g_mutex);
+  if (!g_out) {
+    const char* logdir = getenv("SOFA_LOGDIR");
+    if (!logdir || !*logdir) logdir = ".";
+    char path[4096];
+    snprintf(path, sizeof(path), "%s/rcclshim_%d.sgt", logdir, getpid());
+    g_out = fopen(path, "wb");
+    if (g_out) {
+      sgt::FileHeader hdr{};
+      hdr.magic = sgt::kMagic;
+      hdr.version = sgt::kVersion;
+      hdr.pid = (uint32_t) getpid();
+      hdr.realtime_ns = now_ns(CLOCK_REALTIME);
+      hdr.monotonic_raw_ns = now_ns(CLOCK_MONOTONIC_RAW);
+      // rocp clock not available here; mono_raw stands in (rocp offset 0
+      // flags this file as host-clock-domain for the parser)
+      hdr.rocp_ns = 0;
+      fwrite(&hdr, sizeof(hdr), 1, g_out);
+      sgt::ClockRec cr{};
+      cr.h = {sgt::REC_CLOCK, sizeof(sgt::ClockRec), 0};
+      cr.realtime_ns = hdr.realtime_ns;
+      cr.monotonic_raw_ns = hdr.monotonic_raw_ns;
+      cr.rocp_ns = hdr.monotonic_raw_ns;  // records are stamped in mono_raw
+      fwrite(&cr, sizeof(cr), 1, g_out);
+      // op-name table: ids match the emit() call sites below
+      static const char* names[] = {
+          "ncclAllGather", "ncclAllReduce", "ncclAllToAll", "ncclBroadcast",
+          "ncclReduce", "ncclReduceScatter", "ncclSend", "ncclRecv"};
+      for (uint32_t i = 0; i < 8; ++i) {
+        size_t len = strlen(names[i]);
+        size_t total = (sizeof(sgt::OpNameRec) + len + 1 + 7) & ~size_t(7);
+        char buf[128] = {0};
+        auto* rec = reinterpret_cast<sgt::OpNameRec*>(buf);
+        rec->h = {sgt::REC_OPNAME, (uint16_t) total, 0};
+        rec->kind = 9999;  // shim-private kind; parser matches by name prefix
+        rec->op = i;
+        memcpy(buf + sizeof(sgt::OpNameRec), names[i], len);
+        fwrite(buf, 1, total, g_out);
+      }
+      fflush(g_out);
+    }
+  }
+  pthread_mutex_unlock(&g_mutex);
+}
+
+void emit(uint32_t op, uint64_t t0, uint64_t t1, size_t count, int dt,
+          int peer_or_root, ncclComm_t comm, hipStream_t stream) {
+  ensure_open();
+  if (!g_out) return;
+  sgt::RcclRec r{};
+  r.h = {sgt::REC_RCCL, sizeof(sgt::RcclRec), 0};
+  r.start_ns = t0;
+  r.end_ns = t1;
+  r.tid = (uint32_t) gettid();
+  r.op = op;
+  r.count = count;
+  r.datatype = (uint32_t) dt;
+  r.elem_size = elem_size(dt);
+  r.peer_or_root = peer_or_root;
+  r.device = 0;
+  r.comm = (uint64_t) comm;
+  r.stream = (uint64_t) stream;
+  pthread_mutex_lock(&g_mutex);
+  fwrite(&r, sizeof(r), 1, g_out);
+  pthread_mutex_unlock(&g_mutex);
+}
+
+template <typename Fn>
+Fn next_sym(const char* name) {
+  return reinterpret_cast<Fn>(dlsym(RTLD_NEXT, name));
+}
+
+}  // namespace
+
+#define SHIM(op_id, name, proto, args, count_expr, dt_expr, peer_expr, comm_, \
+             stream_)                                                         \
+  extern "C" ncclResult_t name proto {                                        \
+    using fn_t = ncclResult_t (*) proto;                                      \
+    static fn_t real = next_sym<fn_t>(#name);                                 \
+    if (!real) return 1; /* ncclUnhandledCudaError-ish */                     \
+    uint64_t t0 = now_ns(CLOCK_MONOTONIC_RAW);                                \
+    ncclResult_t rc = real args;                                              \
+    uint64_t t1 = now_ns(CLOCK_MONOTONIC_RAW);                                \
+    emit(op_id, t0, t1, (count_expr), (dt_expr), (peer_expr), (comm_),        \
+         (stream_));                                                          \
+    return rc;                                                                \
+  }
+
+SHIM(0, ncclAllGather,
+     (const void* sb, void* rb, size_t count, ncclDataType_t dt, ncclComm_t comm, hipStream_t st),
+     (sb, rb, count, dt, comm, st), count, dt, -1, comm, st)
+SHIM(1, ncclAllReduce,
+     (const void* sb, void* rb, size_t count, ncclDataType_t dt, ncclRedOp_t op, ncclComm_t comm, hipStream_t st),
+     (sb, rb, count, dt, op, comm, st), count, dt, -1, comm, st)
+SHIM(2, ncclAllToAll,
+     (const void* sb, void* rb, size_t count, ncclDataType_t dt, ncclComm_t comm, hipStream_t st),
+     (sb, rb, count, dt, comm, st), count, dt, -1, comm, st)
+SHIM(3, ncclBroadcast,
+     (const void* sb, void* rb, size_t count, ncclDataType_t dt, int root, ncclComm_t comm, hipStream_t st),
+     (sb, rb, count, dt, root, comm, st), count, dt, root, comm, st)
+SHIM(4, ncclReduce,
+     (const void* sb, void* rb, size_t count, ncclDataType_t dt, ncclRedOp_t op, int root, ncclComm_t comm, hipStream_t st),
+     (sb, rb, count, dt, op, root, comm, st), count, dt, root, comm, st)
+SHIM(5, ncclReduceScatter,
+     (const void* sb, void* rb, size_t recvcount, ncclDataType_t dt, ncclRedOp_t op, ncclComm_t comm, hipStream_t st),
+     (sb, rb, recvcount, dt, op, comm, st), recvcount, dt, -1, comm, st)
+SHIM(6, ncclSend,
+     (const void* sb, size_t count, ncclDataType_t dt, int peer, ncclComm_t comm, hipStream_t st),
+     (sb, count, dt, peer, comm, st), count, dt, peer, comm, st)
+SHIM(7, ncclRecv,
+     (void* rb, size_t count, ncclDataType_t dt, int peer, ncclComm_t comm, hipStream_t st),
+     (rb, count, dt, peer, comm, st), count, dt, peer, comm, st)
+
+__attribute__((destructor)) static void shim_fini() {
+  if (g_out) {
+    fflush(g_out);
+    fclose(g_out);
+    g_out = nullptr;
+  }
+}

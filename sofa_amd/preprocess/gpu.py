@@ -37,7 +37,10 @@ RCCL_COLL_NAMES = None  # resolved from opnames table per file
 
 def load_sgt_files(logdir: str) -> List[SgtFile]:
     out = []
-    for path in sorted(glob.glob(os.path.join(logdir, "gputrace_*.sgt"))):
+    for path in sorted(
+        glob.glob(os.path.join(logdir, "gputrace_*.sgt"))
+        + glob.glob(os.path.join(logdir, "rcclshim_*.sgt"))
+    ):
         try:
             out.append(parse_sgt(path))
         except (ValueError, OSError) as e:

@@ -161,6 +161,13 @@ def build_target_env(cfg: SofaConfig) -> dict:
             env["SOFA_GPU_BUFFER_MB"] = str(cfg.gpu_ring_buffer_mb)
         else:
             p.print_warning("libsofatracer.so not built; GPU tracing disabled")
+        if cfg.rccl_shim:
+            shim = native_lib("libsofarccl.so")
+            if os.path.exists(shim):
+                prev_ld = env.get("LD_PRELOAD", "")
+                env["LD_PRELOAD"] = shim + ((":" + prev_ld) if prev_ld else "")
+            else:
+                p.print_warning("libsofarccl.so not built; RCCL shim disabled")
     return env
 
 
