@@ -394,8 +394,7 @@ void write_opnames() {
   // needs the SDK at analysis time
   for (auto kind : {ROCPROFILER_BUFFER_TRACING_HIP_RUNTIME_API,
                     ROCPROFILER_BUFFER_TRACING_MEMORY_COPY,
-                    ROCPROFILER_BUFFER_TRACING_MEMORY_ALLOCATION,
-                    ROCPROFILER_BUFFER_TRACING_RCCL_API}) {
+                    ROCPROFILER_BUFFER_TRACING_MEMORY_ALLOCATION}) {
     rocprofiler_iterate_buffer_tracing_kind_operations(
         kind,
         [](rocprofiler_buffer_tracing_kind_t k, rocprofiler_tracing_operation_t op,
@@ -411,6 +410,22 @@ void write_opnames() {
         },
         nullptr);
   }
+  // RCCL records come from CALLBACK tracing: its op-name table lives under
+  // the callback kind, not the buffer kind
+  rocprofiler_iterate_callback_tracing_kind_operations(
+      ROCPROFILER_CALLBACK_TRACING_RCCL_API,
+      [](rocprofiler_callback_tracing_kind_t k, rocprofiler_tracing_operation_t op,
+         void*) -> int {
+        const char* name = nullptr;
+        uint64_t len = 0;
+        if (rocprofiler_query_callback_tracing_kind_operation_name(
+                k, op, &name, &len) == ROCPROFILER_STATUS_SUCCESS &&
+            name) {
+          write_opname_rec(1000 + k, op, name);  // offset: callback-kind space
+        }
+        return 1;
+      },
+      nullptr);
 }
 
 int tool_init(rocprofiler_client_finalize_t, void*) {

@@ -32,7 +32,21 @@ COPY_OP_MAP = {
     4: (8, "CopyDeviceToDevice"),
 }
 
-RCCL_COLL_NAMES = None  # resolved from opnames table per file
+# rocprofiler_rccl_api_id_t enum order (rocprofiler-sdk/rccl/api_id.h:34-76)
+# — fallback when a trace carries no RCCL op-name table
+RCCL_API_ID_NAMES = [
+    "ncclAllGather", "ncclAllReduce", "ncclAllToAll", "ncclAllToAllv",
+    "ncclBroadcast", "ncclGather", "ncclReduce", "ncclReduceScatter",
+    "ncclScatter", "ncclSend", "ncclRecv", "ncclRedOpCreatePreMulSum",
+    "ncclRedOpDestroy", "ncclGroupStart", "ncclGroupEnd", "ncclGetVersion",
+    "ncclGetUniqueId", "ncclCommInitRank", "ncclCommInitAll",
+    "ncclCommInitRankConfig", "ncclCommFinalize", "ncclCommDestroy",
+    "ncclCommAbort", "ncclCommSplit", "ncclGetErrorString",
+    "ncclGetLastError", "ncclCommGetAsyncError", "ncclCommCount",
+    "ncclCommCuDevice", "ncclCommUserRank", "ncclMemAlloc", "ncclMemFree",
+    "mscclLoadAlgo", "mscclRunAlgo", "mscclUnloadAlgo", "ncclCommRegister",
+    "ncclCommDeregister", "ncclAllReduceWithBias",
+]
 
 
 def load_sgt_files(logdir: str) -> List[SgtFile]:
@@ -192,11 +206,17 @@ def sgt_to_rccltrace(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.DataFra
         dur = (r["end_ns"] - r["start_ns"]).astype(np.float64) * 1e-9
         payload = (r["count"] * r["elem_size"]).astype(np.int64)
         # opnames stores (kind,op); the RCCL kind enum value varies across SDK
-        # versions, so filter by name prefix instead of kind.
-        rccl_names = {}
-        for (kind, op), nm in sgt.opnames.items():
-            if nm.startswith("nccl") or nm.startswith("mscclpp"):
-                rccl_names[op] = nm
+        # versions, so filter by name prefix; fall back to the embedded
+        # api_id.h table for traces without an op-name dump (e.g. rccl_shim
+        # traces carry their own, collector traces from older builds none).
+        rccl_names = dict(enumerate(RCCL_API_ID_NAMES))
+        shim_names = {
+            op: nm
+            for (kind, op), nm in sgt.opnames.items()
+            if (nm.startswith("nccl") or nm.startswith("msccl"))
+        }
+        if shim_names:
+            rccl_names = shim_names
         op = r["op"]
         df["timestamp"] = ts
         df["duration"] = dur
