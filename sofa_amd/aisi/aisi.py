@@ -97,12 +97,34 @@ def iter_profile(
     }
 
 
-def sofa_aisi(logdir, cfg, df_cpu, df_gpu, df_rccl, features) -> Optional[pd.DataFrame]:
+def tokenize_strace(df_strace: pd.DataFrame):
+    """Syscall stream -> tokens (reference --aisi_via_strace path)."""
+    if df_strace is None or len(df_strace) == 0:
+        return [], np.empty(0), np.empty(0), {}
+    d = df_strace.sort_values("timestamp")
+    base = d["name"].astype(str).str.split("(").str[0]
+    cats = base.astype("category")
+    tokens = cats.cat.codes.to_numpy()
+    ts = d["timestamp"].to_numpy()
+    te = ts + d["duration"].to_numpy()
+    return tokens.tolist(), ts, te, dict(enumerate(cats.cat.categories))
+
+
+def sofa_aisi(logdir, cfg, df_cpu, df_gpu, df_rccl, features,
+              df_strace=None) -> Optional[pd.DataFrame]:
     p.print_title("AISI — iteration detection")
-    if df_gpu is None or len(df_gpu) == 0:
+    if getattr(cfg, "aisi_via_strace", False):
+        tokens, ts, te, vocab = tokenize_strace(df_strace)
+        if not tokens:
+            p.print_warning("no syscall trace for --aisi_via_strace")
+            return None
+        if df_gpu is None:
+            df_gpu = df_strace.iloc[0:0]
+    elif df_gpu is None or len(df_gpu) == 0:
         p.print_warning("no GPU trace; AISI needs kernel records")
         return None
-    tokens, ts, te, vocab = tokenize_kernels(df_gpu)
+    else:
+        tokens, ts, te, vocab = tokenize_kernels(df_gpu)
     if not tokens:
         p.print_warning("no kernels to tokenize")
         return None

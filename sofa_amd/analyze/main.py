@@ -126,7 +126,10 @@ def sofa_analyze(cfg: SofaConfig, pre: Optional[dict] = None) -> Dict[str, float
         try:
             from ..aisi import sofa_aisi
 
-            sofa_aisi(logdir, cfg, df_cpu, df_gpu, df_rccl, features)
+            sofa_aisi(
+                logdir, cfg, df_cpu, df_gpu, df_rccl, features,
+                df_strace=pre.get("df_strace"),
+            )
         except Exception as e:
             p.print_warning(f"AISI failed: {e}")
 

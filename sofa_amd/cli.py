@@ -78,6 +78,7 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--num_swarms", type=int, default=10)
     # analyze
     ap.add_argument("--enable_aisi", action="store_true")
+    ap.add_argument("--aisi_via_strace", action="store_true")
     ap.add_argument("--num_iterations", type=int, default=20)
     ap.add_argument("--spotlight_gpu", action="store_true")
     ap.add_argument("--profile_region", default=None, help="begin,end seconds")
@@ -136,6 +137,7 @@ def cfg_from_args(args) -> SofaConfig:
         enable_swarms=args.enable_swarms,
         num_swarms=args.num_swarms,
         enable_aisi=args.enable_aisi,
+        aisi_via_strace=args.aisi_via_strace,
         num_iterations=args.num_iterations,
         spotlight_gpu=args.spotlight_gpu,
         profile_region=args.profile_region,

@@ -98,3 +98,27 @@ def test_sofa_aisi_end_to_end(tmp_path):
     import os
 
     assert os.path.isfile(os.path.join(str(tmp_path), "iteration_timeline.txt"))
+
+
+def test_aisi_via_strace(tmp_path):
+    """Reference --aisi_via_strace path: iterations from the syscall stream."""
+    block = ["read(ret=4096) 10us", "write(ret=4096) 10us", "pread64(ret=1) 5us", "fsync(ret=0) 2us"]
+    n_iters = 15
+    rows = n_iters * len(block)
+    df = new_trace_df(rows)
+    ts, names = [], []
+    t = 0.0
+    for _ in range(n_iters):
+        for nm in block:
+            ts.append(t)
+            names.append(nm)
+            t += 1e-3
+    df["timestamp"] = ts
+    df["duration"] = 5e-4
+    df["name"] = names
+    cfg = SofaConfig(logdir=str(tmp_path), num_iterations=n_iters)
+    cfg.aisi_via_strace = True
+    feats = []
+    idf = sofa_aisi(str(tmp_path), cfg, None, None, None, feats, df_strace=df)
+    assert idf is not None
+    assert abs(len(idf) - n_iters) <= 2
