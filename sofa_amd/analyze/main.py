@@ -116,6 +116,35 @@ def sofa_analyze(cfg: SofaConfig, pre: Optional[dict] = None) -> Dict[str, float
     comm_mod.comm_profile(logdir, df_gpu, features)
     comm_mod.rccl_link_attribution(logdir, df_rccl, topo, features)
 
+    # --- clock-sync validation (timebase microkernel vs rocprofiler) ---
+    tb_path = os.path.join(logdir, "gpu_timebase.json")
+    sgt_files = pre.get("sgt_files") or []
+    if os.path.isfile(tb_path) and sgt_files:
+        try:
+            with open(tb_path) as f:
+                gtb = json.load(f)
+            for sgt in sgt_files:
+                if len(sgt.clocks) >= 2:
+                    rt0, mono0, rocp0 = sgt.clocks[0]
+                    rt1, mono1, rocp1 = sgt.clocks[-1]
+                    span = mono1 - mono0
+                    if span > 0:
+                        drift_ppm = abs((rocp1 - rocp0) - (mono1 - mono0)) / span * 1e6
+                        features.append(("clock_drift_ppm", float(drift_ppm)))
+                        msg = (
+                            "rocprofiler clock vs CLOCK_MONOTONIC_RAW drift: "
+                            "%.1f ppm over %.2f s; device tick rate %.4f MHz "
+                            "(s_memrealtime microkernel)"
+                            % (drift_ppm, span * 1e-9, gtb.get("ticks_per_second", 0) / 1e6)
+                        )
+                        if drift_ppm > 100:
+                            p.print_warning(msg)
+                        else:
+                            p.print_info(msg)
+                    break
+        except (OSError, ValueError, KeyError):
+            pass
+
     # --- concurrency breakdown ---
     concurrency_breakdown(
         logdir, df_mpstat, df_gpusmi, df_netstat, features, window_s=1.0 / cfg.sys_mon_rate

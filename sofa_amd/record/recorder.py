@@ -188,6 +188,13 @@ def sofa_record(command: str, cfg: SofaConfig) -> int:
         pass
     if cfg.enable_gpu:
         dump_xgmi_topology(logdir)
+        try:
+            from .gpu_timebase import write_gpu_timebase
+
+            if write_gpu_timebase(logdir):
+                p.print_info("GPU timebase microkernel correlation recorded")
+        except Exception as e:
+            p.print_warning(f"gpu timebase prologue failed: {e}")
 
     # --- background monitors ---
     mon = SysMonitor(logdir, rate_hz=cfg.sys_mon_rate, enable_gpu=cfg.enable_gpu)
