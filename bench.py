@@ -139,10 +139,11 @@ def main() -> int:
         step()
     barrier_sync()
 
-    # ---- phase A: plain (no profiler) ----
-    t_plain = timed_phase(args.steps)
-
-    # ---- phase B: profiled (collector armed + SysMonitor + cpusampler) ----
+    # ---- interleaved A/B phases: [plain, profiled] x chunks ----
+    # (a single long A then B pair is vulnerable to clock/thermal drift;
+    # alternating chunks pairs each profiled chunk with an adjacent plain
+    # one — the reference methodology's paired-run idea, in-process)
+    t_plain = 0.0
     t_prof = None
     n_events = 0
     lib = None
@@ -168,16 +169,28 @@ def main() -> int:
                     [sampler_bin, "-o", os.path.join(logdir, f"bench_{rank}.scs"),
                      "-F", "99", "-p", str(os.getpid())]
                 )
-        lib.sofa_tracer_start()
         # profiled warmup: first traced launches pay one-time interception
         # setup; keep that out of the timed region
+        lib.sofa_tracer_start()
         for _ in range(max(2, args.warmup // 2)):
             step()
-        barrier_sync()
-        n0 = lib.sofa_tracer_event_count()
-        t_prof = timed_phase(args.steps)
         lib.sofa_tracer_stop()
+        barrier_sync()
+
+        chunks = 3 if args.steps >= 6 else 1
+        per = max(args.steps // chunks, 1)
+        t_prof = 0.0
+        n0 = lib.sofa_tracer_event_count()
+        done_plain = done_prof = 0
+        for _ in range(chunks):
+            t_plain += timed_phase(per)
+            done_plain += per
+            lib.sofa_tracer_start()
+            t_prof += timed_phase(per)
+            lib.sofa_tracer_stop()
+            done_prof += per
         n_events = int(lib.sofa_tracer_event_count() - n0)
+        args.steps = done_prof  # actual timed steps per mode
         if sampler is not None:
             sampler.terminate()
             sampler.wait(timeout=5)
@@ -185,7 +198,9 @@ def main() -> int:
             mon.terminate()
             mon.wait(timeout=5)
     else:
-        # CPU smoke path: rerun plain so the output shape stays identical
+        # no-profile / CPU smoke path: both phases run plain so the JSON
+        # shape stays identical
+        t_plain = timed_phase(args.steps)
         t_prof = timed_phase(args.steps)
 
     # ---- aggregate across ranks: MAX time, SUM events ----
