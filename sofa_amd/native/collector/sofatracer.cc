@@ -406,7 +406,7 @@ void write_opnames() {
               name) {
             write_opname_rec(k, op, name);
           }
-          return 1;
+          return 0;  // 0 = continue iterating (nonzero stops)
         },
         nullptr);
   }
@@ -423,7 +423,7 @@ void write_opnames() {
             name) {
           write_opname_rec(1000 + k, op, name);  // offset: callback-kind space
         }
-        return 1;
+        return 0;  // 0 = continue iterating
       },
       nullptr);
 }

@@ -209,14 +209,21 @@ def sgt_to_rccltrace(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.DataFra
         # versions, so filter by name prefix; fall back to the embedded
         # api_id.h table for traces without an op-name dump (e.g. rccl_shim
         # traces carry their own, collector traces from older builds none).
-        rccl_names = dict(enumerate(RCCL_API_ID_NAMES))
-        shim_names = {
-            op: nm
-            for (kind, op), nm in sgt.opnames.items()
-            if (nm.startswith("nccl") or nm.startswith("msccl"))
-        }
-        if shim_names:
-            rccl_names = shim_names
+        # shim traces use a private id space marked kind==9999; collector
+        # traces use SDK api ids — start from the embedded api_id.h table and
+        # let any discovered names (same id space) override
+        shim_table = {op: nm for (kind, op), nm in sgt.opnames.items() if kind == 9999}
+        if shim_table:
+            rccl_names = shim_table
+        else:
+            rccl_names = dict(enumerate(RCCL_API_ID_NAMES))
+            rccl_names.update(
+                {
+                    op: nm
+                    for (kind, op), nm in sgt.opnames.items()
+                    if (nm.startswith("nccl") or nm.startswith("msccl"))
+                }
+            )
         op = r["op"]
         df["timestamp"] = ts
         df["duration"] = dur
