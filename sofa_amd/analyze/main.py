@@ -112,7 +112,7 @@ def sofa_analyze(cfg: SofaConfig, pre: Optional[dict] = None) -> Dict[str, float
     profiles.netbandwidth_profile(logdir, features)
     profiles.net_profile(logdir, df_net, features)
     profiles.gpu_profile(df_gpu, df_rccl, features)
-    profiles.gpusmi_profile(df_gpusmi, features)
+    profiles.gpusmi_profile(df_gpusmi, features, logdir)
     comm_mod.comm_profile(logdir, df_gpu, features)
     comm_mod.rccl_link_attribution(logdir, df_rccl, topo, features)
 

@@ -91,7 +91,7 @@ def gpu_profile(df_gpu: pd.DataFrame, df_rccl: pd.DataFrame, features: Features)
             print("    %8.4f s %6d  %s" % (row["sum"], int(row["count"]), short))
 
 
-def gpusmi_profile(df_sm: pd.DataFrame, features: Features) -> None:
+def gpusmi_profile(df_sm: pd.DataFrame, features: Features, logdir: str = "") -> None:
     if df_sm is None or len(df_sm) == 0:
         return
     print("\nGPU utilization (rocm-smi):")
@@ -100,6 +100,22 @@ def gpusmi_profile(df_sm: pd.DataFrame, features: Features) -> None:
         print("  gpu%-2d busy%%: q25=%.0f q50=%.0f q75=%.0f max=%.0f" % (dev, q25, q50, q75, q100))
         features.append((f"gpu{dev}_util_q50", float(q50)))
         features.append((f"gpu{dev}_util_max", float(q100)))
+    # VRAM/power from the telemetry CSV (wider than the trace series)
+    path = os.path.join(logdir, "gpusmi_trace.csv") if logdir else ""
+    if path and os.path.isfile(path):
+        try:
+            d = pd.read_csv(path)
+            if "vram_MB" in d:
+                for dev, grp in d.groupby("dev"):
+                    features.append((f"gpu{dev}_vram_peak_MB", float(grp["vram_MB"].max())))
+                    if "power_W" in grp:
+                        features.append((f"gpu{dev}_power_mean_W", float(grp["power_W"].mean())))
+                print(
+                    "  VRAM peak %.0f MB; mean power %.0f W (per-GPU detail in features.csv)"
+                    % (d["vram_MB"].max(), d.get("power_W", pd.Series([0])).mean())
+                )
+        except (OSError, ValueError, KeyError):
+            pass
 
 
 def mpstat_profile(df_mp: pd.DataFrame, features: Features, idle_threshold: float = 10.0) -> None:

@@ -9,6 +9,11 @@
 // against any librccl.
 //
 // Enable:  LD_PRELOAD=/path/libsofarccl.so SOFA_LOGDIR=<logdir> <cmd>
+//
+// LIMITATION (measured): PyTorch-ROCm bundles its own librccl and resolves
+// nccl* internally in a way LD_PRELOAD does not interpose — for torch use
+// the rocprofiler-sdk path (default).  The shim covers apps that link RCCL
+// normally (rccl-tests, custom C++/MPI jobs).
 
 #include <dlfcn.h>
 #include <pthread.h>

@@ -241,4 +241,6 @@ def parse_gpusmi(logdir: str, tb: Optional[TimeBase]):
 
     csv = d.copy()
     csv["timestamp"] = _tl(tb, d["ts"])
+    csv["vram_MB"] = d["vram"].clip(lower=0) / 1e6
+    csv["power_W"] = d["power"].clip(lower=0)
     return sm, mem, csv
