@@ -69,3 +69,30 @@ def test_swarm_diff(tmp_path):
     alpha = out[out["caption"].str.contains("alpha")]
     assert len(alpha) == 1 and alpha["match_cluster"].iloc[0] >= 0
     assert os.path.isfile(os.path.join(str(tmp_path), "swarm_diff.csv"))
+
+
+def test_diff_cli_verb(tmp_path):
+    """`sofa diff` end-to-end through the CLI (with --skip_preprocess)."""
+    import subprocess
+    import sys
+
+    base_dir = tmp_path / "base"
+    match_dir = tmp_path / "match"
+    base_dir.mkdir()
+    match_dir.mkdir()
+    hsg_cluster(_synth_cpu(["alpha @ x.so", "beta @ y.so"]), 2, str(base_dir))
+    hsg_cluster(_synth_cpu(["alpha @ x.so", "beta @ y.so"]), 2, str(match_dir))
+    sofa = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "bin", "sofa")
+    r = subprocess.run(
+        [
+            sys.executable, sofa, "diff",
+            "--base_logdir", str(base_dir),
+            "--match_logdir", str(match_dir),
+            "--logdir", str(tmp_path),
+            "--skip_preprocess",
+        ],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert r.returncode == 0, r.stderr
+    assert "intersection rate" in r.stdout
+    assert os.path.isfile(os.path.join(str(tmp_path), "swarm_diff.csv"))
