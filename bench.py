@@ -61,6 +61,8 @@ def main() -> int:
                     "the reference's opt-in --cuda_api_tracing)")
     ap.add_argument("--full-record", type=int, default=1,
                     help="also run SysMonitor + cpusampler during profiled phase")
+    ap.add_argument("--sampler", type=int, default=1, help="cpusampler on/off within full-record")
+    ap.add_argument("--monitor", type=int, default=1, help="SysMonitor on/off within full-record")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -157,14 +159,14 @@ def main() -> int:
         if args.full_record:
             # telemetry monitor as a SUBPROCESS (the real `sofa record`
             # architecture: pollers live in the recorder, not the target)
-            if local_rank == 0:
+            if local_rank == 0 and args.monitor:
                 mon = subprocess.Popen(
                     [sys.executable, "-m", "sofa_amd.record.monitor_main",
                      "--logdir", logdir, "--rate", "10", "--parent", str(os.getpid())],
                     cwd=REPO,
                 )
             sampler_bin = os.path.join(REPO, "sofa_amd", "native", "bin", "sofa-cpusampler")
-            if os.path.exists(sampler_bin):
+            if os.path.exists(sampler_bin) and args.sampler:
                 sampler = subprocess.Popen(
                     [sampler_bin, "-o", os.path.join(logdir, f"bench_{rank}.scs"),
                      "-F", "99", "-p", str(os.getpid())]
