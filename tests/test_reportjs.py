@@ -12,7 +12,7 @@ SOFA = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
 def test_report_js_structure(tmp_path, native_built):
     logdir = str(tmp_path / "log")
     r = subprocess.run(
-        [sys.executable, SOFA, "stat", "dd if=/dev/zero of=/dev/null bs=1M count=400",
+        [sys.executable, SOFA, "stat", "dd if=/dev/zero of=/dev/null bs=1M count=8000",
          "--logdir", logdir, "--no_gpu"],
         capture_output=True, text=True, timeout=120,
     )
