@@ -52,6 +52,8 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--enable_tcpdump", action="store_true")
     ap.add_argument("--enable_strace", action="store_true")
     ap.add_argument("--enable_py_stacks", action="store_true")
+    ap.add_argument("--call_stacks", action="store_true",
+                    help="sample native call stacks (-g) -> flamegraph.folded + flame.html")
     ap.add_argument("--no_gpu", action="store_true", help="disable GPU tracing")
     ap.add_argument(
         "--hip_api_trace",
@@ -124,6 +126,7 @@ def cfg_from_args(args) -> SofaConfig:
         enable_tcpdump=args.enable_tcpdump,
         enable_strace=args.enable_strace,
         enable_pystacks=args.enable_py_stacks,
+        enable_callchain=args.call_stacks,
         enable_gpu=not args.no_gpu,
         enable_gpu_hip_api=args.hip_api_trace,
         enable_rccl_trace=not args.no_rccl,

@@ -75,6 +75,7 @@ class SofaConfig:
     enable_tcpdump: bool = False       # AF_PACKET sniffer
     enable_strace: bool = False
     enable_pystacks: bool = False
+    enable_callchain: bool = False   # -g native call stacks -> flamegraph
     enable_vmstat: bool = True
     enable_diskstat: bool = True
     enable_netstat: bool = True

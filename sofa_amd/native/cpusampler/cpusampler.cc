@@ -62,7 +62,10 @@ enum RecType : uint16_t {
   REC_COMM = 3,
   REC_EXIT = 4,
   REC_LOST = 5,
+  REC_SAMPLE_CS = 6,  // sample + fixed-depth callchain (-g mode)
 };
+
+constexpr int kMaxFrames = 16;
 
 struct RecHeader {
   uint16_t type;
@@ -77,6 +80,19 @@ struct SampleRec {  // REC_SAMPLE
   uint32_t cpu;
   uint32_t flags;  // bit0: kernel-space sample
   uint64_t period;
+};
+
+struct SampleCsRec {  // REC_SAMPLE_CS: fixed size for vectorized parsing
+  RecHeader h;
+  uint64_t time_ns;
+  uint64_t ip;
+  uint32_t pid, tid;
+  uint32_t cpu;
+  uint32_t flags;
+  uint64_t period;
+  uint32_t n_frames;
+  uint32_t _pad;
+  uint64_t frames[kMaxFrames];  // user-space return addresses, leaf first
 };
 
 struct MmapRec {  // REC_MMAP, followed by filename (size-derived length)
@@ -150,7 +166,8 @@ struct Ring {
 };
 
 // perf sample layout given our sample_type =
-// IP | TID | TIME | CPU | PERIOD (in that order per ABI).
+// IP | TID | TIME | CPU | PERIOD (in that order per ABI); with -g,
+// CALLCHAIN (u64 nr; u64 ips[nr]) follows PERIOD.
 struct RawSample {
   uint64_t ip;
   uint32_t pid, tid;
@@ -158,6 +175,8 @@ struct RawSample {
   uint32_t cpu, res;
   uint64_t period;
 };
+
+bool g_callchain = false;
 
 void drain_ring(Ring& r, Writer& w) {
   auto* m = r.meta;
@@ -184,19 +203,44 @@ void drain_ring(Ring& r, Writer& w) {
     switch (ph->type) {
       case PERF_RECORD_SAMPLE: {
         const auto* s = reinterpret_cast<const RawSample*>(body);
-        SampleRec rec{};
-        rec.h = {REC_SAMPLE, sizeof(SampleRec)};
-        rec.time_ns = s->time;
-        rec.ip = s->ip;
-        rec.pid = s->pid;
-        rec.tid = s->tid;
-        rec.cpu = s->cpu;
-        rec.flags = (ph->misc & PERF_RECORD_MISC_CPUMODE_MASK) ==
-                            PERF_RECORD_MISC_KERNEL
-                        ? 1u
-                        : 0u;
-        rec.period = s->period;
-        w.append(&rec, sizeof(rec));
+        uint32_t kflag = (ph->misc & PERF_RECORD_MISC_CPUMODE_MASK) ==
+                                 PERF_RECORD_MISC_KERNEL
+                             ? 1u
+                             : 0u;
+        if (!g_callchain) {
+          SampleRec rec{};
+          rec.h = {REC_SAMPLE, sizeof(SampleRec)};
+          rec.time_ns = s->time;
+          rec.ip = s->ip;
+          rec.pid = s->pid;
+          rec.tid = s->tid;
+          rec.cpu = s->cpu;
+          rec.flags = kflag;
+          rec.period = s->period;
+          w.append(&rec, sizeof(rec));
+        } else {
+          SampleCsRec rec{};
+          rec.h = {REC_SAMPLE_CS, sizeof(SampleCsRec)};
+          rec.time_ns = s->time;
+          rec.ip = s->ip;
+          rec.pid = s->pid;
+          rec.tid = s->tid;
+          rec.cpu = s->cpu;
+          rec.flags = kflag;
+          rec.period = s->period;
+          const char* cc = body + sizeof(RawSample);
+          uint64_t nr = 0;
+          memcpy(&nr, cc, sizeof(nr));
+          const uint64_t* ips = reinterpret_cast<const uint64_t*>(cc + 8);
+          uint32_t out_n = 0;
+          for (uint64_t f = 0; f < nr && out_n < kMaxFrames; ++f) {
+            // skip PERF_CONTEXT_* markers ((u64)-128 .. (u64)-4096 sentinels)
+            if (ips[f] >= 0xfffffffffffff000ull) continue;
+            rec.frames[out_n++] = ips[f];
+          }
+          rec.n_frames = out_n;
+          w.append(&rec, sizeof(rec));
+        }
         break;
       }
       case PERF_RECORD_MMAP2: {
@@ -284,6 +328,7 @@ int main(int argc, char** argv) {
     else if (a == "-p" && i + 1 < argc) target_pid = atoi(argv[++i]);
     else if (a == "-a") system_wide = true;
     else if (a == "--max-mb" && i + 1 < argc) max_mb = strtoull(argv[++i], nullptr, 10);
+    else if (a == "-g" || a == "--callchain") g_callchain = true;
     else {
       fprintf(stderr, "usage: %s -o out.scs [-F hz] (-a | -p pid) [--max-mb N]\n", argv[0]);
       return 2;
@@ -337,6 +382,10 @@ int main(int argc, char** argv) {
   attr.use_clockid = 1;
   attr.clockid = CLOCK_MONOTONIC_RAW;
   attr.wakeup_events = 16;
+  if (g_callchain) {
+    attr.sample_type |= PERF_SAMPLE_CALLCHAIN;
+    attr.sample_max_stack = 32;
+  }
 
   std::vector<Ring> rings;
   const size_t n_pages = 64;  // 64 data pages = 256 KiB per CPU

@@ -90,6 +90,15 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
     else:
         p.print_warning("no CPU samples")
     result["df_cpu"] = df_cpu
+    if scs is not None and len(scs.samples_cs):
+        try:
+            from .flame import write_folded
+
+            out = write_folded(scs, logdir)
+            if out:
+                p.print_info(f"flamegraph input written: {out}")
+        except Exception as e:
+            p.print_warning(f"stack folding failed: {e}")
 
     # ---------------- system monitors ----------------
     t_mp, mp_csv, usr_sys = sysmon.parse_mpstat(logdir, tb)

@@ -24,6 +24,9 @@ REC_MMAP = 2
 REC_COMM = 3
 REC_EXIT = 4
 REC_LOST = 5
+REC_SAMPLE_CS = 6
+
+MAX_FRAMES = 16
 
 SAMPLE_SIZE = 48
 # RecHeader(4) + pad(4 by alignment)?  cpusampler.cc SampleRec:
@@ -45,6 +48,26 @@ SAMPLE_DTYPE = np.dtype(
 )
 assert SAMPLE_DTYPE.itemsize == SAMPLE_SIZE
 
+# -g mode: fixed-depth callchain samples (cpusampler.cc SampleCsRec, 184 B)
+SAMPLE_CS_DTYPE = np.dtype(
+    [
+        ("type", "<u2"),
+        ("size", "<u2"),
+        ("_pad", "<u4"),
+        ("time_ns", "<u8"),
+        ("ip", "<u8"),
+        ("pid", "<u4"),
+        ("tid", "<u4"),
+        ("cpu", "<u4"),
+        ("flags", "<u4"),
+        ("period", "<u8"),
+        ("n_frames", "<u4"),
+        ("_pad2", "<u4"),
+        ("frames", "<u8", (MAX_FRAMES,)),
+    ]
+)
+assert SAMPLE_CS_DTYPE.itemsize == 184
+
 MMAP_FIXED = 40  # RecHeader(4)+pad(4)+time(8)+pid(4)+tid(4)+addr(8)+len(8)+pgoff(8) = 48?
 
 
@@ -56,6 +79,7 @@ class ScsFile:
     sample_freq: int = 0
     n_cpus: int = 0
     samples: np.ndarray = field(default_factory=lambda: np.empty(0, dtype=SAMPLE_DTYPE))
+    samples_cs: np.ndarray = field(default_factory=lambda: np.empty(0, dtype=SAMPLE_CS_DTYPE))
     # pid -> list of (addr, len, pgoff, filename)
     mmaps: Dict[int, List[Tuple[int, int, int, str]]] = field(default_factory=dict)
     comms: Dict[int, str] = field(default_factory=dict)
@@ -78,6 +102,7 @@ def parse_scs(path: str) -> ScsFile:
     out.n_cpus = ncpus
 
     sample_chunks: List[np.ndarray] = []
+    cs_chunks: List[np.ndarray] = []
     off = HEADER_SIZE
     n = len(buf)
     u16 = np.frombuffer(buf[: n & ~1], dtype="<u2")
@@ -104,6 +129,18 @@ def parse_scs(path: str) -> ScsFile:
             arr = np.frombuffer(buf, dtype=SAMPLE_DTYPE, count=k, offset=off)
             sample_chunks.append(arr)
             off += k * SAMPLE_SIZE
+        elif rtype == REC_SAMPLE_CS and rsize == SAMPLE_CS_DTYPE.itemsize:
+            item = SAMPLE_CS_DTYPE.itemsize
+            max_k = (n - off) // item
+            base = off // 2
+            stride = item // 2
+            types = u16[base : base + max_k * stride : stride]
+            sizes = u16[base + 1 : base + 1 + max_k * stride : stride]
+            bad = np.nonzero((types != REC_SAMPLE_CS) | (sizes != item))[0]
+            k = int(bad[0]) if len(bad) else max_k
+            arr = np.frombuffer(buf, dtype=SAMPLE_CS_DTYPE, count=k, offset=off)
+            cs_chunks.append(arr)
+            off += k * item
         elif rtype == REC_MMAP:
             (time_ns, pid, tid, addr, ln, pgoff) = struct.unpack_from("<QIIQQQ", buf, off + 8)
             name = buf[off + 8 + 40 : off + rsize].split(b"\0", 1)[0].decode("utf-8", "replace")
@@ -122,4 +159,15 @@ def parse_scs(path: str) -> ScsFile:
             off += rsize
     if sample_chunks:
         out.samples = np.concatenate(sample_chunks) if len(sample_chunks) > 1 else sample_chunks[0]
+    if cs_chunks:
+        out.samples_cs = np.concatenate(cs_chunks) if len(cs_chunks) > 1 else cs_chunks[0]
+        # expose CS samples through the flat view too (shared field prefix)
+        flat = np.zeros(len(out.samples_cs), dtype=SAMPLE_DTYPE)
+        for f in ("time_ns", "ip", "pid", "tid", "cpu", "flags", "period"):
+            flat[f] = out.samples_cs[f]
+        flat["type"] = REC_SAMPLE
+        flat["size"] = SAMPLE_SIZE
+        out.samples = (
+            np.concatenate([out.samples, flat]) if len(out.samples) else flat
+        )
     return out

@@ -245,6 +245,8 @@ def sofa_record(command: str, cfg: SofaConfig) -> int:
     sampler_bin = native_bin("sofa-cpusampler")
     if os.path.exists(sampler_bin):
         args = [sampler_bin, "-o", os.path.join(logdir, "cpusamples.scs"), "-F", str(cfg.cpu_sample_rate)]
+        if cfg.enable_callchain:
+            args.append("-g")
         if cfg.profile_all_cpus:
             args.append("-a")
         else:
