@@ -29,7 +29,7 @@ def concurrency_breakdown(
     t0 = df_mpstat["timestamp"].min()
     t1 = df_mpstat["timestamp"].max()
     if t1 <= t0:
-        return None
+        t1 = t0 + window_s  # sub-period run: one window still gets attributed
     edges = np.arange(t0, t1 + window_s, window_s)
     n_win = len(edges) - 1
     if n_win <= 0:
