@@ -13,7 +13,6 @@ import os
 import shutil
 from typing import Dict, List, Optional, Tuple
 
-import numpy as np
 import pandas as pd
 
 from .. import printing as p

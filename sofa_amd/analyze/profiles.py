@@ -10,7 +10,7 @@ Parity map (reference bin/sofa_analyze.py):
 from __future__ import annotations
 
 import os
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import numpy as np
 import pandas as pd

@@ -10,7 +10,6 @@ intersection rate, write swarm_diff.csv.
 from __future__ import annotations
 
 import os
-from typing import Dict
 
 import pandas as pd
 

@@ -15,7 +15,6 @@ from typing import Optional
 import numpy as np
 import pandas as pd
 
-from ..config import SofaConfig
 from ..schema import new_trace_df
 from .scs import ScsFile, parse_scs
 from .symbols import Symbolizer

@@ -12,12 +12,11 @@ from __future__ import annotations
 
 import glob
 import os
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import numpy as np
 import pandas as pd
 
-from ..config import SofaConfig
 from ..schema import new_trace_df, trace_df_from
 from .sgt import SgtFile, parse_sgt
 from .symbols import demangle

@@ -9,7 +9,7 @@ SURVEY.md §4).
 from __future__ import annotations
 
 import os
-from typing import List, Optional
+from typing import List
 
 import numpy as np
 import pandas as pd
@@ -20,7 +20,7 @@ from ..schema import SOFATrace, new_trace_df, traces_to_json, write_trace_csv
 from . import cpu as cpu_mod
 from . import gpu as gpu_mod
 from . import sysmon
-from .timebase import TimeBase, load_timebase
+from .timebase import load_timebase
 
 FILTER_COLORS = [
     "red", "orange", "yellow", "green", "blue", "indigo", "violet",

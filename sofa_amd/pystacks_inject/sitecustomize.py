@@ -15,7 +15,6 @@ pystacks.txt shape parsed at bin/sofa_preprocess.py:1709-1761):
 import os
 
 if os.environ.get("SOFA_PYSTACKS_OUT"):
-    import atexit
     import sys
     import threading
     import time

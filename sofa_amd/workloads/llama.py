@@ -14,7 +14,6 @@ Run directly for a profiled step stress:
 from __future__ import annotations
 
 import argparse
-import math
 
 import torch
 import torch.nn as nn
