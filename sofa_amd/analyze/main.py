@@ -28,9 +28,16 @@ def _load_csv_trace(logdir: str, name: str) -> pd.DataFrame:
     if not os.path.isfile(path):
         return new_trace_df(0)
     try:
-        return pd.read_csv(path)
-    except (OSError, ValueError):
+        df = pd.read_csv(path)
+    except Exception:
+        p.print_warning(f"{name} unreadable; ignoring")
         return new_trace_df(0)
+    # a corrupt/foreign CSV must not crash the profiles downstream
+    required = {"timestamp", "duration", "name", "copyKind", "deviceId"}
+    if not required <= set(df.columns):
+        p.print_warning(f"{name} lacks trace columns; ignoring")
+        return new_trace_df(0)
+    return df
 
 
 def _roi_filter(df: pd.DataFrame, begin: float, end: float) -> pd.DataFrame:
