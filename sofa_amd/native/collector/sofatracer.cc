@@ -44,6 +44,7 @@
 #include <ctime>
 #include <mutex>
 #include <string>
+#include <string_view>
 #include <unordered_map>
 #include <vector>
 
@@ -478,9 +479,46 @@ int tool_init(rocprofiler_client_finalize_t, void*) {
   rocprofiler_configure_buffer_tracing_service(
       g_ctx, ROCPROFILER_BUFFER_TRACING_MEMORY_COPY, nullptr, 0, g_buffer);
   if (env_flag("SOFA_TRACE_HIP_API", true)) {
-    rocprofiler_configure_buffer_tracing_service(
-        g_ctx, ROCPROFILER_BUFFER_TRACING_HIP_RUNTIME_API, nullptr, 0,
-        g_buffer);
+    // SOFA_HIP_API_OPS: "all" = every HIP call (expensive at launch rates),
+    // default = launches/copies/syncs/allocs only (the calls a timeline
+    // reader actually drills into; cuts record volume massively)
+    const char* ops_sel = getenv("SOFA_HIP_API_OPS");
+    bool all_ops = ops_sel && strcmp(ops_sel, "all") == 0;
+    if (all_ops) {
+      rocprofiler_configure_buffer_tracing_service(
+          g_ctx, ROCPROFILER_BUFFER_TRACING_HIP_RUNTIME_API, nullptr, 0,
+          g_buffer);
+    } else {
+      static std::vector<rocprofiler_tracing_operation_t> ops;
+      rocprofiler_iterate_buffer_tracing_kind_operations(
+          ROCPROFILER_BUFFER_TRACING_HIP_RUNTIME_API,
+          [](rocprofiler_buffer_tracing_kind_t k,
+             rocprofiler_tracing_operation_t op, void*) -> int {
+            const char* name = nullptr;
+            uint64_t len = 0;
+            if (rocprofiler_query_buffer_tracing_kind_operation_name(
+                    k, op, &name, &len) == ROCPROFILER_STATUS_SUCCESS &&
+                name) {
+              std::string_view n{name};
+              if (n.find("Launch") != std::string_view::npos ||
+                  n.find("Memcpy") != std::string_view::npos ||
+                  n.find("Memset") != std::string_view::npos ||
+                  n.find("Synchronize") != std::string_view::npos ||
+                  n.find("Malloc") != std::string_view::npos ||
+                  n.find("hipFree") != std::string_view::npos ||
+                  n.find("GraphLaunch") != std::string_view::npos ||
+                  n.find("EventRecord") != std::string_view::npos ||
+                  n.find("StreamWaitEvent") != std::string_view::npos) {
+                ops.push_back(op);
+              }
+            }
+            return 0;
+          },
+          nullptr);
+      rocprofiler_configure_buffer_tracing_service(
+          g_ctx, ROCPROFILER_BUFFER_TRACING_HIP_RUNTIME_API, ops.data(),
+          ops.size(), g_buffer);
+    }
   }
   if (env_flag("SOFA_TRACE_ALLOC", false)) {
     rocprofiler_configure_buffer_tracing_service(
