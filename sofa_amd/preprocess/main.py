@@ -213,8 +213,16 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
         p.print_warning(f"pystacks parse failed: {e}")
         result["df_pystacks"] = new_trace_df(0)
 
-    # ---------------- report.js ----------------
+    # ---------------- report.js + chrome trace ----------------
     traces_to_json(traces, os.path.join(logdir, "report.js"), plot_ratio=cfg.plot_ratio)
+    try:
+        from ..viz.chrome_trace import write_chrome_trace
+
+        ct = write_chrome_trace(logdir, result)
+        if ct:
+            p.print_info(f"chrome://tracing export: {ct}")
+    except Exception as e:
+        p.print_warning(f"chrome trace export failed: {e}")
     n_events = sum(len(t.data) for t in traces if t.data is not None)
     p.print_progress(f"preprocess done: {len(traces)} series, {n_events} points")
     result["traces"] = traces

@@ -35,3 +35,26 @@ def test_report_js_structure(tmp_path, native_built):
     listed = [s.strip() for s in m.group(1).split(",")]
     assert set(listed) <= set(names)
     assert "cpu_traces" in listed
+
+
+def test_chrome_trace_export(tmp_path):
+    """chrome_trace.json loads as valid trace-event JSON."""
+    import numpy as np
+
+    from sofa_amd.schema import new_trace_df, write_trace_csv
+    from sofa_amd.viz.chrome_trace import write_chrome_trace
+
+    df = new_trace_df(4)
+    df["timestamp"] = [0.1, 0.2, 0.3, 0.4]
+    df["duration"] = 1e-3
+    df["deviceId"] = [0, 0, 1, 1]
+    df["copyKind"] = [0, 1, 0, 16]
+    df["name"] = ["k1", "copy", "k2", "ncclAllReduce"]
+    write_trace_csv(df, os.path.join(tmp_path, "gputrace.csv"))
+    out = write_chrome_trace(str(tmp_path))
+    assert out
+    data = json.load(open(out))
+    evs = data["traceEvents"]
+    assert len(evs) == 3  # 2 kernels + 1 copy (rccl csv separate)
+    assert all(e["ph"] == "X" and e["dur"] > 0 for e in evs)
+    assert {e["pid"] for e in evs} == {"GPU kernels", "GPU copies"}
