@@ -55,6 +55,6 @@ def test_chrome_trace_export(tmp_path):
     assert out
     data = json.load(open(out))
     evs = data["traceEvents"]
-    assert len(evs) == 3  # 2 kernels + 1 copy (rccl csv separate)
+    assert len(evs) == 4  # 2 kernels + 2 non-kernel rows
     assert all(e["ph"] == "X" and e["dur"] > 0 for e in evs)
     assert {e["pid"] for e in evs} == {"GPU kernels", "GPU copies"}
