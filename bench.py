@@ -56,9 +56,9 @@ def main() -> int:
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--no-profile", action="store_true", help="skip the profiled phase")
-    ap.add_argument("--hip-api", type=int, default=0,
-                    help="trace HIP runtime API (0/1; off by default, matching "
-                    "the reference's opt-in --cuda_api_tracing)")
+    ap.add_argument("--hip-api", type=int, default=1,
+                    help="trace HIP runtime API spans (filtered op set; "
+                    "SOFA_HIP_API_OPS=all for every call)")
     ap.add_argument("--full-record", type=int, default=1,
                     help="also run SysMonitor + cpusampler during profiled phase")
     ap.add_argument("--sampler", type=int, default=1, help="cpusampler on/off within full-record")

@@ -55,11 +55,13 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--call_stacks", action="store_true",
                     help="sample native call stacks (-g) -> flamegraph.folded + flame.html")
     ap.add_argument("--no_gpu", action="store_true", help="disable GPU tracing")
+    ap.add_argument("--no_hip_api", action="store_true",
+                    help="disable HIP runtime API span tracing")
     ap.add_argument(
-        "--hip_api_trace",
+        "--hip_api_full",
         action="store_true",
-        help="also trace HIP runtime API spans (opt-in; adds ~45%% overhead "
-        "on launch-dense workloads)",
+        help="trace EVERY HIP API call, not just launch/copy/sync/alloc "
+        "(adds ~45%% overhead on launch-dense workloads)",
     )
     ap.add_argument("--no_rccl", action="store_true")
     ap.add_argument(
@@ -128,7 +130,8 @@ def cfg_from_args(args) -> SofaConfig:
         enable_pystacks=args.enable_py_stacks,
         enable_callchain=args.call_stacks,
         enable_gpu=not args.no_gpu,
-        enable_gpu_hip_api=args.hip_api_trace,
+        enable_gpu_hip_api=not args.no_hip_api,
+        hip_api_full=args.hip_api_full,
         enable_rccl_trace=not args.no_rccl,
         rccl_shim=args.rccl_shim,
         gpu_ring_buffer_mb=args.gpu_buffer_mb,

@@ -80,10 +80,12 @@ class SofaConfig:
     enable_diskstat: bool = True
     enable_netstat: bool = True
     enable_gpu: bool = True            # rocprofiler-sdk collector
-    # HIP runtime API spans are OPT-IN (reference parity: --cuda_api_tracing
-    # was opt-in too, bin/sofa:*) — measured +~45% overhead on launch-dense
-    # training steps (profiles/overhead_matrix notes)
-    enable_gpu_hip_api: bool = False
+    # HIP runtime API spans: default ON with the FILTERED op set (launches/
+    # copies/syncs/allocs — measured ~0 extra overhead at 2x events/s);
+    # full API tracing (--hip_api_full) costs ~45% on launch-dense steps
+    # (profiles/overhead_matrix_r01.md)
+    enable_gpu_hip_api: bool = True
+    hip_api_full: bool = False
     enable_rccl_trace: bool = True     # RCCL API tracing via collector
     rccl_shim: bool = False            # LD_PRELOAD interposer (fallback path)
     enable_kfd_trace: bool = False     # page-migrate/fault events

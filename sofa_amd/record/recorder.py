@@ -157,6 +157,8 @@ def build_target_env(cfg: SofaConfig) -> dict:
             env["ROCP_TOOL_LIBRARIES"] = tracer + ((":" + prev) if prev else "")
             env["SOFA_LOGDIR"] = os.path.abspath(cfg.logdir)
             env["SOFA_TRACE_HIP_API"] = "1" if cfg.enable_gpu_hip_api else "0"
+            if getattr(cfg, "hip_api_full", False):
+                env["SOFA_HIP_API_OPS"] = "all"
             env["SOFA_TRACE_RCCL"] = "1" if cfg.enable_rccl_trace else "0"
             env["SOFA_GPU_BUFFER_MB"] = str(cfg.gpu_ring_buffer_mb)
         else:
