@@ -122,6 +122,14 @@ def sofa_analyze(cfg: SofaConfig, pre: Optional[dict] = None) -> Dict[str, float
     comm_mod.comm_profile(logdir, df_gpu, features)
     comm_mod.rccl_link_attribution(logdir, df_rccl, topo, features)
 
+    # --- launch-latency / launch-bound analysis (corr-id join) ---
+    try:
+        from .launch import launch_latency_profile
+
+        launch_latency_profile(pre.get("sgt_files") or [], features)
+    except Exception as e:
+        p.print_warning(f"launch-latency analysis failed: {e}")
+
     # --- clock-sync validation (timebase microkernel vs rocprofiler) ---
     tb_path = os.path.join(logdir, "gpu_timebase.json")
     sgt_files = pre.get("sgt_files") or []

@@ -139,3 +139,30 @@ def test_concurrency_breakdown_gpu_dominant(tmp_path):
     d = dict(feats)
     assert d["dominant_gpu_ratio"] > 0.9
     assert os.path.isfile(os.path.join(tmp_path, "performance.csv"))
+
+
+def test_launch_latency_profile(tmp_path):
+    """corr-id join of kernels with their launch API spans (synthetic SGT)."""
+    import sys, os
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from sgt_synth import SgtWriter
+    from sofa_amd.preprocess.sgt import parse_sgt
+    from sofa_amd.analyze.launch import launch_latency_profile
+
+    w = SgtWriter()
+    # api span ends at 1000, kernel starts 6000 -> latency 5 us
+    for i in range(10):
+        base = i * 100_000
+        w.hip_api(start=base + 100, end=base + 1000, op=5)
+        # patch corr ids: SgtWriter.hip_api leaves corr 0; write kernels with corr 0 too
+        w.kernel(start=base + 6000, end=base + 9000, kid=1, corr=0)
+    path = tmp_path / "gputrace_1.sgt"
+    w.write(str(path))
+    s = parse_sgt(str(path))
+    feats = []
+    launch_latency_profile([s], feats)
+    d = dict(feats)
+    assert "launch_latency_us_p50" in d
+    assert 4.0 <= d["launch_latency_us_p50"] <= 6.0
+    # 3us busy vs 91us gaps -> launch-bound ratio high
+    assert d["gpu_idle_gap_ratio"] > 0.5
