@@ -92,12 +92,13 @@ class SgtWriter:
         a["bytes"] = nbytes
         self.buf += a.tobytes()
 
-    def hip_api(self, start, end, op, tid=1):
+    def hip_api(self, start, end, op, tid=1, corr=0):
         a = np.zeros(1, API_DTYPE)
         a["type"] = REC_HIPAPI
         a["size"] = API_DTYPE.itemsize
         a["start_ns"] = start
         a["end_ns"] = end
+        a["corr_id"] = corr
         a["tid"] = tid
         a["op"] = op
         self.buf += a.tobytes()

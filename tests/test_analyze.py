@@ -150,12 +150,11 @@ def test_launch_latency_profile(tmp_path):
     from sofa_amd.analyze.launch import launch_latency_profile
 
     w = SgtWriter()
-    # api span ends at 1000, kernel starts 6000 -> latency 5 us
+    # api span ends at 1000, kernel starts 6000 -> latency 5 us per launch
     for i in range(10):
         base = i * 100_000
-        w.hip_api(start=base + 100, end=base + 1000, op=5)
-        # patch corr ids: SgtWriter.hip_api leaves corr 0; write kernels with corr 0 too
-        w.kernel(start=base + 6000, end=base + 9000, kid=1, corr=0)
+        w.hip_api(start=base + 100, end=base + 1000, op=5, corr=i + 1)
+        w.kernel(start=base + 6000, end=base + 9000, kid=1, corr=i + 1)
     path = tmp_path / "gputrace_1.sgt"
     w.write(str(path))
     s = parse_sgt(str(path))
