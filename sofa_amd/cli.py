@@ -71,6 +71,8 @@ def build_parser() -> argparse.ArgumentParser:
         "the rocprofiler-sdk path",
     )
     ap.add_argument("--gpu_buffer_mb", type=int, default=64)
+    ap.add_argument("--enable_kfd_trace", action="store_true",
+                    help="trace KFD page-migrate/fault events (SVM memory pressure)")
     # preprocess
     ap.add_argument("--cpu_time_offset_ms", type=int, default=0)
     ap.add_argument("--plot_ratio", type=int, default=1)
@@ -135,6 +137,7 @@ def cfg_from_args(args) -> SofaConfig:
         enable_rccl_trace=not args.no_rccl,
         rccl_shim=args.rccl_shim,
         gpu_ring_buffer_mb=args.gpu_buffer_mb,
+        enable_kfd_trace=args.enable_kfd_trace,
         cpu_time_offset_ms=args.cpu_time_offset_ms,
         plot_ratio=args.plot_ratio,
         cpu_filters=parse_filters(args.cpu_filters),

@@ -37,6 +37,15 @@ enum RecType : uint16_t {
   REC_ALLOC = 9,        // AllocRec
   REC_DROP = 10,        // DropRec (buffer drops, should be 0 for lossless)
   REC_MARKER = 11,      // NameRec-shaped: roctx range/instant (id -> message)
+  REC_KFD = 12,         // KfdRec: page migrate/fault/queue events
+};
+
+// KFD event classes (KfdRec.op_class)
+enum KfdClass : uint32_t {
+  KFD_PAGE_MIGRATE = 1,
+  KFD_PAGE_FAULT = 2,
+  KFD_QUEUE_EVT = 3,
+  KFD_UNMAP = 4,
 };
 
 struct RecHeader {
@@ -147,5 +156,18 @@ struct DropRec {
   RecHeader h;  // REC_DROP
   uint64_t dropped;
 };
+
+struct KfdRec {
+  RecHeader h;        // REC_KFD
+  uint64_t timestamp; // KFD-reported ns (rocprofiler clock domain)
+  uint32_t op_class;  // KfdClass
+  uint32_t operation; // start/end/etc per class
+  uint32_t pid;
+  int32_t device;     // logical GPU (or -1)
+  uint64_t addr_start;
+  uint64_t addr_end;
+  int32_t src_device; // migrate: source agent's logical id
+  int32_t error_code;
+};  // 56 bytes
 
 }  // namespace sgt

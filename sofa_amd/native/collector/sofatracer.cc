@@ -332,6 +332,41 @@ void buffer_callback(rocprofiler_context_id_t, rocprofiler_buffer_id_t,
         emit(&rec, sizeof(rec));
         break;
       }
+      case ROCPROFILER_BUFFER_TRACING_KFD_EVENT_PAGE_MIGRATE: {
+        auto* r = static_cast<
+            rocprofiler_buffer_tracing_kfd_event_page_migrate_record_t*>(
+            header->payload);
+        sgt::KfdRec rec{};
+        rec.h = {sgt::REC_KFD, sizeof(sgt::KfdRec), 0};
+        rec.timestamp = r->timestamp;
+        rec.op_class = sgt::KFD_PAGE_MIGRATE;
+        rec.operation = r->operation;
+        rec.pid = r->pid;
+        rec.device = agent_device(r->dst_agent);
+        rec.src_device = agent_device(r->src_agent);
+        rec.addr_start = r->start_address.value;
+        rec.addr_end = r->end_address.value;
+        rec.error_code = r->error_code;
+        emit(&rec, sizeof(rec));
+        break;
+      }
+      case ROCPROFILER_BUFFER_TRACING_KFD_EVENT_PAGE_FAULT: {
+        auto* r = static_cast<
+            rocprofiler_buffer_tracing_kfd_event_page_fault_record_t*>(
+            header->payload);
+        sgt::KfdRec rec{};
+        rec.h = {sgt::REC_KFD, sizeof(sgt::KfdRec), 0};
+        rec.timestamp = r->timestamp;
+        rec.op_class = sgt::KFD_PAGE_FAULT;
+        rec.operation = r->operation;
+        rec.pid = r->pid;
+        rec.device = agent_device(r->agent_id);
+        rec.src_device = -1;
+        rec.addr_start = r->address.value;
+        rec.addr_end = r->address.value;
+        emit(&rec, sizeof(rec));
+        break;
+      }
       case ROCPROFILER_BUFFER_TRACING_MEMORY_ALLOCATION: {
         auto* r =
             static_cast<rocprofiler_buffer_tracing_memory_allocation_record_t*>(
@@ -523,6 +558,15 @@ int tool_init(rocprofiler_client_finalize_t, void*) {
   if (env_flag("SOFA_TRACE_ALLOC", false)) {
     rocprofiler_configure_buffer_tracing_service(
         g_ctx, ROCPROFILER_BUFFER_TRACING_MEMORY_ALLOCATION, nullptr, 0,
+        g_buffer);
+  }
+  if (env_flag("SOFA_TRACE_KFD", false)) {
+    // page-migrate/fault events (SVM memory pressure diagnosis)
+    rocprofiler_configure_buffer_tracing_service(
+        g_ctx, ROCPROFILER_BUFFER_TRACING_KFD_EVENT_PAGE_MIGRATE, nullptr, 0,
+        g_buffer);
+    rocprofiler_configure_buffer_tracing_service(
+        g_ctx, ROCPROFILER_BUFFER_TRACING_KFD_EVENT_PAGE_FAULT, nullptr, 0,
         g_buffer);
   }
   if (env_flag("SOFA_TRACE_RCCL", true)) {

@@ -286,3 +286,44 @@ def sgt_to_hip_api_trace(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.Dat
     out.sort_values("timestamp", inplace=True, kind="stable")
     out.reset_index(drop=True, inplace=True)
     return out
+
+
+KFD_CLASS_NAMES = {1: "page_migrate", 2: "page_fault", 3: "queue_evt", 4: "unmap"}
+
+
+def sgt_to_kfdtrace(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.DataFrame:
+    """KFD page-migrate/fault events -> unified rows (category 4)."""
+    frames = []
+    for sgt in files:
+        k = sgt.kfd
+        if not len(k):
+            continue
+        nbytes = (k["addr_end"] - k["addr_start"]).astype(np.int64)
+        cls = k["op_class"]
+        names = np.array(
+            [
+                "kfd_%s op%d gpu%d %d bytes"
+                % (KFD_CLASS_NAMES.get(int(c), str(c)), o, max(d, 0), b)
+                for c, o, d, b in zip(cls, k["operation"], k["device"], nbytes)
+            ],
+            dtype=object,
+        )
+        df = trace_df_from(
+            len(k),
+            timestamp=_timeline(tb, sgt, k["timestamp"]),
+            duration=np.full(len(k), 1e-6),
+            deviceId=k["device"].astype(np.int64),
+            payload=np.maximum(nbytes, 0),
+            pid=k["pid"].astype(np.int64),
+            event=cls.astype(np.float64),
+            pkt_src=k["src_device"].astype(np.int64),
+            name=names,
+            category=np.full(len(k), 4, dtype=np.int64),
+        )
+        frames.append(df)
+    if not frames:
+        return new_trace_df(0)
+    out = pd.concat(frames, ignore_index=True)
+    out.sort_values("timestamp", inplace=True, kind="stable")
+    out.reset_index(drop=True, inplace=True)
+    return out

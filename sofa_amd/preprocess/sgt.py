@@ -30,6 +30,7 @@ REC_CLOCK = 8
 REC_ALLOC = 9
 REC_DROP = 10
 REC_MARKER = 11
+REC_KFD = 12
 
 _HDR = [("type", "<u2"), ("size", "<u2"), ("_pad", "<u4")]
 
@@ -99,6 +100,21 @@ RCCL_DTYPE = np.dtype(
     ]
 )
 
+KFD_DTYPE = np.dtype(
+    _HDR
+    + [
+        ("timestamp", "<u8"),
+        ("op_class", "<u4"),
+        ("operation", "<u4"),
+        ("pid", "<u4"),
+        ("device", "<i4"),
+        ("addr_start", "<u8"),
+        ("addr_end", "<u8"),
+        ("src_device", "<i4"),
+        ("error_code", "<i4"),
+    ]
+)
+
 ALLOC_DTYPE = np.dtype(
     _HDR
     + [
@@ -116,6 +132,7 @@ ALLOC_DTYPE = np.dtype(
 
 FIXED_DTYPES = {
     REC_KERNEL: KERNEL_DTYPE,
+    REC_KFD: KFD_DTYPE,
     REC_COPY: COPY_DTYPE,
     REC_HIPAPI: API_DTYPE,
     REC_RCCL: RCCL_DTYPE,
@@ -134,6 +151,7 @@ class SgtFile:
     hip_api: np.ndarray = field(default_factory=lambda: np.empty(0, API_DTYPE))
     rccl: np.ndarray = field(default_factory=lambda: np.empty(0, RCCL_DTYPE))
     allocs: np.ndarray = field(default_factory=lambda: np.empty(0, ALLOC_DTYPE))
+    kfd: np.ndarray = field(default_factory=lambda: np.empty(0, KFD_DTYPE))
     kernel_names: Dict[int, str] = field(default_factory=dict)
     markers: List[tuple] = field(default_factory=list)  # (rocp_ns, message)
     opnames: Dict[tuple, str] = field(default_factory=dict)  # (kind, op) -> name
@@ -255,4 +273,5 @@ def parse_sgt(path: str) -> SgtFile:
     out.hip_api = cat(REC_HIPAPI, API_DTYPE)
     out.rccl = cat(REC_RCCL, RCCL_DTYPE)
     out.allocs = cat(REC_ALLOC, ALLOC_DTYPE)
+    out.kfd = cat(REC_KFD, KFD_DTYPE)
     return out

@@ -161,6 +161,8 @@ def build_target_env(cfg: SofaConfig) -> dict:
                 env["SOFA_HIP_API_OPS"] = "all"
             env["SOFA_TRACE_RCCL"] = "1" if cfg.enable_rccl_trace else "0"
             env["SOFA_GPU_BUFFER_MB"] = str(cfg.gpu_ring_buffer_mb)
+            if cfg.enable_kfd_trace:
+                env["SOFA_TRACE_KFD"] = "1"
         else:
             p.print_warning("libsofatracer.so not built; GPU tracing disabled")
         if cfg.rccl_shim:
