@@ -251,7 +251,10 @@ def sofa_record(command: str, cfg: SofaConfig) -> int:
         args = [sampler_bin, "-o", os.path.join(logdir, "cpusamples.scs"), "-F", str(cfg.cpu_sample_rate)]
         if cfg.enable_callchain:
             args.append("-g")
-        if cfg.profile_all_cpus:
+        if cfg.profile_all_cpus or cfg.enable_strace:
+            # strace mode: the real workload is a grandchild the ptrace
+            # wrapper forked before the sampler attached, so per-pid inherit
+            # would miss it — sample system-wide instead
             args.append("-a")
         else:
             args += ["-p", str(target.pid)]
