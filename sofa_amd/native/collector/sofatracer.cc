@@ -75,7 +75,11 @@ uint64_t host_ns(clockid_t c) {
   return uint64_t(ts.tv_sec) * 1000000000ull + ts.tv_nsec;
 }
 
+bool g_null_sink = false;  // SOFA_NULL_SINK=1: intercept but discard (the
+                           // SDK-interception floor for overhead attribution)
+
 void write_raw(const void* p, size_t n) {
+  if (g_null_sink) return;
   std::lock_guard<std::mutex> lk(g_mutex);
   if (g_out) fwrite(p, 1, n, g_out);
 }
@@ -465,6 +469,7 @@ void write_opnames() {
 }
 
 int tool_init(rocprofiler_client_finalize_t, void*) {
+  g_null_sink = env_flag("SOFA_NULL_SINK", false);
   const char* logdir = getenv("SOFA_LOGDIR");
   if (!logdir || !*logdir) logdir = ".";
   char path[4096];
