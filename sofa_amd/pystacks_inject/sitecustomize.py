@@ -24,7 +24,9 @@ if os.environ.get("SOFA_PYSTACKS_OUT"):
 
     def _sampler():
         try:
-            f = open(_out_path + ".%d" % os.getpid(), "w")
+            # line-buffered: the daemon thread dies with the process, so
+            # anything still in a block buffer would be lost
+            f = open(_out_path + ".%d" % os.getpid(), "w", buffering=1)
         except OSError:
             return
         period = 1.0 / max(_hz, 1.0)

@@ -71,3 +71,30 @@ def test_clean(tmp_path, native_built):
 def test_record_needs_command():
     r = run_sofa(["record"])
     assert r.returncode == 2
+
+
+def test_pystacks_e2e(tmp_path, native_built):
+    logdir = str(tmp_path / "log")
+    code = "import time\nt=time.time()\nx=0\nwhile time.time()-t<1.5: x+=1\n"
+    script = tmp_path / "busy.py"
+    script.write_text(code)
+    r = run_sofa(
+        ["stat", f"{sys.executable} {script}", "--logdir", logdir, "--no_gpu", "--enable_py_stacks"]
+    )
+    assert "Complete!!" in r.stdout, r.stderr
+    import glob
+
+    assert glob.glob(os.path.join(logdir, "pystacks.txt.*")), "no pystacks raw output"
+    assert os.path.isfile(os.path.join(logdir, "pystacks.csv"))
+
+
+def test_cluster_report_cli(tmp_path, native_built):
+    """`sofa report --cluster_ip a,b` over two recorded per-node logdirs."""
+    base = str(tmp_path / "clog")
+    for ip in ("10.0.0.1", "10.0.0.2"):
+        r = run_sofa(["stat", "sleep 0.6", "--logdir", f"{base}-{ip}", "--no_gpu"])
+        assert "Complete!!" in r.stdout
+    r = run_sofa(["report", "--logdir", base, "--cluster_ip", "10.0.0.1,10.0.0.2", "--no_gpu"])
+    assert r.returncode == 0, r.stderr
+    assert "Complete!!" in r.stdout
+    assert os.path.isfile(os.path.join(base, "cluster_report.csv"))
