@@ -24,7 +24,6 @@ DDP:  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
 
 import argparse
 import ctypes
-import glob
 import json
 import os
 import sys
@@ -102,7 +101,6 @@ def main() -> int:
             bucket_cap_mb=64,  # fewer, larger RCCL buckets for per-link xGMI efficiency
         )
     opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
-    scaler = None
     bs = args.batch
     x = torch.randn(bs, 3, 224, 224, device=device)
     if use_cuda:
