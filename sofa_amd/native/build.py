@@ -103,6 +103,7 @@ def build_all(verbose: bool = False) -> None:
             [
                 os.path.join(HERE, "hip", "timebase_kernel.hip"),
                 os.path.join(HERE, "hip", "trace_ring.hip"),
+                os.path.join(HERE, "hip", "ring_writer.hip"),
             ],
         ),
         (
