@@ -49,6 +49,10 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--cpu_sample_rate", type=int, default=99)
     ap.add_argument("--sys_mon_rate", type=int, default=10)
     ap.add_argument("--profile_all_cpus", action="store_true")
+    ap.add_argument("--attach", type=int, default=0, metavar="PID",
+                    help="record an already-running process instead of launching one")
+    ap.add_argument("--duration", type=float, default=0.0,
+                    help="with --attach: stop after N seconds")
     ap.add_argument("--enable_tcpdump", action="store_true")
     ap.add_argument("--enable_strace", action="store_true")
     ap.add_argument("--enable_py_stacks", action="store_true")
@@ -196,10 +200,15 @@ def main(argv=None) -> int:
         return 0
 
     if verb in ("record", "stat"):
-        if not args.usr_command:
-            p.print_error("record/stat needs a command to profile")
+        if args.attach:
+            from .record.recorder import sofa_attach
+
+            sofa_attach(args.attach, cfg, duration=args.duration)
+        elif not args.usr_command:
+            p.print_error("record/stat needs a command to profile (or --attach PID)")
             return 2
-        sofa_record(args.usr_command, cfg)
+        else:
+            sofa_record(args.usr_command, cfg)
         if verb == "record":
             return 0
 

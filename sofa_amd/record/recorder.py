@@ -175,6 +175,73 @@ def build_target_env(cfg: SofaConfig) -> dict:
     return env
 
 
+def sofa_attach(pid: int, cfg: SofaConfig, duration: float = 0.0) -> int:
+    """Observe an already-running process: CPU sampling + system monitors
+    (the GPU collector cannot attach post-init — rocprofiler tools load with
+    the runtime — so GPU activity comes from telemetry only in this mode).
+
+    Runs until the target exits, `duration` elapses, or Ctrl-C.
+    """
+    logdir = ensure_logdir(cfg)
+    ensure_native_built(cfg.verbose)
+    sofa_clean(cfg)
+    if not os.path.exists(f"/proc/{pid}"):
+        p.print_error(f"pid {pid} does not exist")
+        return 2
+    p.print_progress(f"attaching to pid {pid} (logdir {logdir})")
+
+    with open(os.path.join(logdir, "sofa_time.txt"), "w") as f:
+        f.write("%.9f\n" % time.time())
+    _write_timebase(logdir)
+    try:
+        shutil.copyfile("/proc/kallsyms", os.path.join(logdir, "kallsyms"))
+    except OSError:
+        pass
+    if cfg.enable_gpu:
+        dump_xgmi_topology(logdir)
+
+    mon = SysMonitor(logdir, rate_hz=cfg.sys_mon_rate, enable_gpu=cfg.enable_gpu)
+    mon.start()
+    sampler = None
+    sampler_bin = native_bin("sofa-cpusampler")
+    if os.path.exists(sampler_bin):
+        args = [sampler_bin, "-o", os.path.join(logdir, "cpusamples.scs"),
+                "-F", str(cfg.cpu_sample_rate), "-p", str(pid)]
+        if cfg.enable_callchain:
+            args.append("-g")
+        sampler = subprocess.Popen(args)
+
+    t_begin = time.time()
+    ret = 0
+    try:
+        while os.path.exists(f"/proc/{pid}"):
+            time.sleep(0.2)
+            if duration and time.time() - t_begin >= duration:
+                break
+    except KeyboardInterrupt:
+        pass
+    t_end = time.time()
+
+    if sampler is not None:
+        try:
+            sampler.terminate()
+            sampler.wait(timeout=5)
+        except (OSError, subprocess.TimeoutExpired):
+            sampler.kill()
+    mon.stop()
+    mon.join(timeout=5)
+    with open(os.path.join(logdir, "misc.txt"), "w") as f:
+        f.write(json.dumps({
+            "elapsed_time": t_end - t_begin,
+            "cores": os.cpu_count(),
+            "pid": pid,
+            "returncode": 0,
+            "command": f"--attach {pid}",
+        }))
+    p.print_progress("attach recording done (%.2f s)" % (t_end - t_begin))
+    return ret
+
+
 def sofa_record(command: str, cfg: SofaConfig) -> int:
     logdir = ensure_logdir(cfg)
     ensure_native_built(cfg.verbose)

@@ -98,3 +98,19 @@ def test_cluster_report_cli(tmp_path, native_built):
     assert r.returncode == 0, r.stderr
     assert "Complete!!" in r.stdout
     assert os.path.isfile(os.path.join(base, "cluster_report.csv"))
+
+
+def test_attach_mode(tmp_path, native_built):
+    """`sofa record --attach PID` observes a running process."""
+    import subprocess as sp
+
+    logdir = str(tmp_path / "log")
+    busy = sp.Popen([sys.executable, "-c",
+                     "import time\nt=time.time()\nx=0\nwhile time.time()-t<3: x+=1"])
+    try:
+        r = run_sofa(["stat", "--attach", str(busy.pid), "--duration", "1.5",
+                      "--logdir", logdir, "--no_gpu"])
+        assert "Complete!!" in r.stdout, (r.stdout[-1500:], r.stderr[-800:])
+        assert os.path.isfile(os.path.join(logdir, "cputrace.csv"))
+    finally:
+        busy.wait()
