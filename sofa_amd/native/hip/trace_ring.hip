@@ -102,35 +102,42 @@ __global__ void compact_kernel(const RingRec* __restrict__ slots, uint32_t n_val
 
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
-  // thread t owns records [base + t*R, base + t*R + R): consecutive per
-  // thread so the kept-record order equals input order after the scan.
-  uint32_t first = (blockIdx.x * kBlockThreads + threadIdx.x) * kRecsPerThread;
+  // wave w owns the contiguous span [span, span + 512); within the span,
+  // lane l's j-th record is span + j*64 + l — COALESCED loads (for fixed j
+  // the 64 lanes read 64 consecutive 32-B records = one contiguous 2 KiB
+  // region), unlike thread-consecutive ownership (256-B lane stride).
+  // Record order within the wave is (j, lane)-major, so stability falls out
+  // of a per-j ballot prefix + a running per-wave total.
+  uint32_t span = (blockIdx.x * kWavesPerBlock + wave) * 64 * kRecsPerThread;
 
   RingRec r[kRecsPerThread];
   bool keep[kRecsPerThread];
-  uint32_t my_count = 0;
 #pragma unroll
   for (int j = 0; j < kRecsPerThread; ++j) {
-    uint32_t i = first + j;
+    uint32_t i = span + j * 64 + lane;
     keep[j] = false;
     if (i < n_valid) {
       r[j] = slots[i];
       keep[j] = r[j].tag != 0 && ((tag_mask >> (r[j].tag & 63u)) & 1ull);
     }
-    my_count += keep[j] ? 1u : 0u;
   }
 
-  // wave-level exclusive scan of per-lane counts (log2(64) shuffle steps)
-  uint32_t scan = my_count;
+  // per-j ballots: output offset of (j, lane) = sum of totals of rounds
+  // < j + count of kept lanes < lane in round j
+  unsigned long long below = (1ull << lane) - 1ull;
+  uint32_t round_base[kRecsPerThread];
+  uint32_t lane_prefix[kRecsPerThread];
+  uint32_t running = 0;
 #pragma unroll
-  for (int d = 1; d < 64; d <<= 1) {
-    uint32_t up = __shfl_up(scan, d, 64);
-    if (lane >= d) scan += up;
+  for (int j = 0; j < kRecsPerThread; ++j) {
+    unsigned long long m = __ballot(keep[j]);
+    round_base[j] = running;
+    lane_prefix[j] = __popcll(m & below);
+    running += __popcll(m);
   }
-  uint32_t lane_base = scan - my_count;               // exclusive
-  uint32_t wave_total = __shfl(scan, 63, 64);         // inclusive of lane 63
+  uint32_t wave_total = running;
 
-  if (lane == 63) wave_totals[wave] = wave_total;
+  if (lane == 0) wave_totals[wave] = wave_total;
   __syncthreads();
 
   if (wave == 0 && lane == 0) {
@@ -143,15 +150,17 @@ __global__ void compact_kernel(const RingRec* __restrict__ slots, uint32_t n_val
   }
   __syncthreads();
 
-  uint32_t dst = block_base + wave_bases[wave] + lane_base;
+  uint32_t base = block_base + wave_bases[wave];
 #pragma unroll
   for (int j = 0; j < kRecsPerThread; ++j) {
-    if (keep[j] && dst < out_cap) {
-      RingRec o = r[j];
-      o.t_start = (uint64_t)((double) o.t_start * scale + (double) offset);
-      o.t_end = (uint64_t)((double) o.t_end * scale + (double) offset);
-      out[dst] = o;
-      ++dst;
+    if (keep[j]) {
+      uint32_t dst = base + round_base[j] + lane_prefix[j];
+      if (dst < out_cap) {
+        RingRec o = r[j];
+        o.t_start = (uint64_t)((double) o.t_start * scale + (double) offset);
+        o.t_end = (uint64_t)((double) o.t_end * scale + (double) offset);
+        out[dst] = o;
+      }
     }
   }
 }
