@@ -159,6 +159,11 @@ def rccl_link_attribution(
     """
     if df_rccl is None or len(df_rccl) == 0:
         return None
+    # collective data ops only: setup calls (ncclCommInitRank & co. — seconds
+    # long, zero payload) must not pollute link time/bandwidth
+    df_rccl = df_rccl[df_rccl["payload"] > 0]
+    if len(df_rccl) == 0:
+        return None
     n = topo.get("n_gpus", 0) if topo else 0
     if n < 2:
         # single-gpu or no topology: report aggregate only
@@ -206,7 +211,7 @@ def rccl_link_attribution(
             q = int(peers[i])
             if 0 <= q < n:
                 add(q, dev, S, d)
-        else:
+        elif S > 0:  # unknown data op: conservative ring-successor estimate
             add(dev, nxt, S, d)
 
     rows = []
