@@ -185,8 +185,21 @@ void drain_ring(Ring& r, Writer& w) {
   const size_t mask = r.data_size - 1;
   std::vector<char> tmp;
   while (tail < head) {
-    auto* eh = reinterpret_cast<struct perf_event_header*>(r.data + (tail & mask));
+    // the 8-byte record header itself may wrap the ring boundary
+    struct perf_event_header hdr_copy;
+    const struct perf_event_header* eh;
+    size_t hpos = tail & mask;
+    if (hpos + sizeof(hdr_copy) > r.data_size) {
+      size_t first = r.data_size - hpos;
+      memcpy(&hdr_copy, r.data + hpos, first);
+      memcpy(reinterpret_cast<char*>(&hdr_copy) + first, r.data,
+             sizeof(hdr_copy) - first);
+      eh = &hdr_copy;
+    } else {
+      eh = reinterpret_cast<struct perf_event_header*>(r.data + hpos);
+    }
     uint16_t esz = eh->size;
+    if (esz == 0) break;  // corrupt ring: bail rather than spin
     const char* ev;
     if (((tail & mask) + esz) > r.data_size) {
       // wrapped event: copy into tmp
