@@ -327,3 +327,26 @@ def sgt_to_kfdtrace(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.DataFram
     out.sort_values("timestamp", inplace=True, kind="stable")
     out.reset_index(drop=True, inplace=True)
     return out
+
+
+def sgt_to_markers(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.DataFrame:
+    """roctx marks/range-pushes -> instant rows (category 5)."""
+    rows = []
+    for sgt in files:
+        if not sgt.markers:
+            continue
+        off = sgt.rocp_to_realtime_offset()
+        for rocp_ns, msg in sgt.markers:
+            ts = (rocp_ns + off) * 1e-9
+            rows.append((ts - (tb.time_base if tb else 0.0), msg, sgt.pid))
+    if not rows:
+        return new_trace_df(0)
+    df = new_trace_df(len(rows))
+    df["timestamp"] = [r[0] for r in rows]
+    df["duration"] = 1e-6
+    df["name"] = ["roctx:" + r[1] for r in rows]
+    df["pid"] = [r[2] for r in rows]
+    df["category"] = 5
+    df.sort_values("timestamp", inplace=True, kind="stable")
+    df.reset_index(drop=True, inplace=True)
+    return df

@@ -168,6 +168,11 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
     if len(df_hip):
         write_trace_csv(df_hip, os.path.join(logdir, "hip_api_trace.csv"))
         traces.append(SOFATrace(name="hip_api_traces", title="HIP API", color="MediumSeaGreen", data=df_hip))
+    df_marks = gpu_mod.sgt_to_markers(sgt_files, tb) if sgt_files else new_trace_df(0)
+    if len(df_marks):
+        write_trace_csv(df_marks, os.path.join(logdir, "markers.csv"))
+        traces.append(SOFATrace(name="roctx_traces", title="roctx markers", color="Black", data=df_marks))
+    result["df_markers"] = df_marks
     df_kfd = gpu_mod.sgt_to_kfdtrace(sgt_files, tb) if sgt_files else new_trace_df(0)
     if len(df_kfd):
         write_trace_csv(df_kfd, os.path.join(logdir, "kfdtrace.csv"))

@@ -57,6 +57,13 @@ class SgtWriter:
             total - 16 - len(nm)
         )
 
+    def marker(self, rocp_ns, msg):
+        nm = msg.encode() + b"\0"
+        total = (16 + len(nm) + 7) & ~7
+        self.buf += struct.pack("<HHIQ", 11, total, 0, rocp_ns) + nm + b"\0" * (
+            total - 16 - len(nm)
+        )
+
     def opname(self, kind, op, name):
         nm = name.encode() + b"\0"
         total = (16 + len(nm) + 7) & ~7

@@ -96,3 +96,20 @@ def test_truncated_sgt_is_tolerated(synth, tmp_path):
     trunc.write_bytes(buf[: len(buf) - 37])  # cut mid-record
     s = parse_sgt(str(trunc))
     assert len(s.kernels) >= 1  # earlier records still parse
+
+
+def test_markers_to_timeline(tmp_path):
+    from sofa_amd.preprocess.gpu import sgt_to_markers
+
+    w = SgtWriter(pid=5, realtime_ns=2_000_000_000_000, rocp_ns=0)
+    w.clock(realtime_ns=2_000_000_000_000, mono=0, rocp=0)
+    w.marker(500_000, "step_begin")
+    w.marker(900_000, "step_end")
+    path = tmp_path / "gputrace_5.sgt"
+    w.write(str(path))
+    s = parse_sgt(str(path))
+    assert len(s.markers) == 2
+    df = sgt_to_markers([s], None)
+    assert len(df) == 2
+    assert df["name"].tolist() == ["roctx:step_begin", "roctx:step_end"]
+    assert abs(df["timestamp"].iloc[0] - 2000.0005) < 1e-6
