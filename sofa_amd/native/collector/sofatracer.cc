@@ -34,6 +34,7 @@
 #include <rocprofiler-sdk/rccl.h>
 #include <rocprofiler-sdk/marker/api_id.h>
 
+#include <dlfcn.h>
 #include <unistd.h>
 
 #include <atomic>
@@ -149,6 +150,20 @@ void code_object_callback(rocprofiler_callback_tracing_record_t record,
   }
 }
 
+// Current HIP device of the calling thread, resolved lazily via dlsym so the
+// tool library carries no link-time HIP dependency (it loads during runtime
+// registration, before HIP is ready; the RCCL callback fires much later).
+// Without this, every rank's collectives would report device 0 and per-link
+// attribution in 8-GPU DDP (ranks pick GPUs via hipSetDevice, not
+// HIP_VISIBLE_DEVICES) would pile onto one ring edge.
+int current_hip_device() {
+  using fn_t = int (*)(int*);
+  static fn_t fn = reinterpret_cast<fn_t>(dlsym(RTLD_DEFAULT, "hipGetDevice"));
+  int dev = 0;
+  if (fn && fn(&dev) == 0) return dev;
+  return 0;
+}
+
 // RCCL callback tracing: start/stop phases; we record the span + args.
 struct RcclPending {
   uint64_t start_ns;
@@ -176,7 +191,7 @@ void rccl_callback(rocprofiler_callback_tracing_record_t record,
   rec.tid = static_cast<uint32_t>(record.thread_id);
   rec.op = record.operation;
   rec.peer_or_root = -1;
-  rec.device = 0;
+  rec.device = (uint32_t) current_hip_device();
 
   const auto& a = data->args;
   switch (record.operation) {
