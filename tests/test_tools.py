@@ -1,0 +1,63 @@
+"""Unit tests for tools/ (xring report aggregation, doctor, etl bench)."""
+
+import importlib.util
+import json
+import os
+import subprocess
+import sys
+
+import pandas as pd
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _load(name, path):
+    spec = importlib.util.spec_from_file_location(name, path)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    return mod
+
+
+def test_xring_report(tmp_path, monkeypatch):
+    xring = _load("xring", os.path.join(REPO, "tools", "xring.py"))
+    base = str(tmp_path / "xringlog")
+    for n, payload in [(2, 1e6), (3, 2e6)]:
+        d = f"{base}-{n}"
+        os.makedirs(d)
+        pd.DataFrame(
+            {"name": ["rccl_payload", "rccl_time", "rccl_hot_link_bytes"],
+             "value": [payload, 0.5, payload / n]}
+        ).to_csv(os.path.join(d, "features.csv"), index=False)
+        pd.DataFrame(
+            {"src": [0], "dst": [1], "bytes": [payload], "time_s": [0.5],
+             "est_bw_GBps": [payload / 0.5 / 1e9]}
+        ).to_csv(os.path.join(d, "xlink_traffic.csv"), index=False)
+    monkeypatch.chdir(tmp_path)
+
+    class A:
+        logdir_base = base
+
+    xring.report(A())
+    out = pd.read_csv(tmp_path / "xring.csv")
+    assert len(out) == 2
+    assert out["n_gpus"].tolist() == [2, 3]
+    assert out["rccl_payload"].tolist() == [1e6, 2e6]
+
+
+def test_doctor_runs():
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "doctor.py")],
+        capture_output=True, text=True, timeout=300,
+    )
+    assert "ROCm (hipcc)" in r.stdout
+    assert "native libsofatracer.so" in r.stdout
+
+
+def test_etl_bench_small():
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "etl_bench.py"), "--n", "20000"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+    )
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert "end-to-end ETL" in r.stdout
