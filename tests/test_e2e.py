@@ -114,3 +114,14 @@ def test_attach_mode(tmp_path, native_built):
         assert os.path.isfile(os.path.join(logdir, "cputrace.csv"))
     finally:
         busy.wait()
+
+
+def test_demo_logdir_analyzes():
+    """The committed demo logdir stays analyzable (guards the demo artifact)."""
+    demo = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "demo", "logdir")
+    if not os.path.isdir(demo):
+        import pytest
+
+        pytest.skip("demo not present")
+    r = run_sofa(["analyze", "--logdir", demo, "--skip_preprocess", "--no_gpu"])
+    assert "Complete!!" in r.stdout
