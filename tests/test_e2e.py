@@ -116,12 +116,17 @@ def test_attach_mode(tmp_path, native_built):
         busy.wait()
 
 
-def test_demo_logdir_analyzes():
-    """The committed demo logdir stays analyzable (guards the demo artifact)."""
+def test_demo_logdir_analyzes(tmp_path):
+    """The committed demo logdir stays analyzable (guards the demo artifact);
+    runs on a COPY so the committed files are never mutated."""
+    import shutil
+
     demo = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "demo", "logdir")
     if not os.path.isdir(demo):
         import pytest
 
         pytest.skip("demo not present")
-    r = run_sofa(["analyze", "--logdir", demo, "--skip_preprocess", "--no_gpu"])
+    work = str(tmp_path / "logdir")
+    shutil.copytree(demo, work)
+    r = run_sofa(["analyze", "--logdir", work, "--skip_preprocess", "--no_gpu"])
     assert "Complete!!" in r.stdout
