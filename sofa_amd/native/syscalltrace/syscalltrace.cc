@@ -149,6 +149,11 @@ int main(int argc, char** argv) {
       ptrace(PTRACE_SYSCALL, tid, nullptr, nullptr);
     } else if (sig == SIGTRAP || event != 0) {
       ptrace(PTRACE_SYSCALL, tid, nullptr, nullptr);
+    } else if (sig == SIGSTOP && pending.find(tid) == pending.end()) {
+      // a new tracee's initial SIGSTOP: suppress it (re-injecting would
+      // group-stop the thread), start syscall-tracing it
+      pending[tid];  // mark seen
+      ptrace(PTRACE_SYSCALL, tid, nullptr, nullptr);
     } else {
       // deliver the real signal
       ptrace(PTRACE_SYSCALL, tid, nullptr, (void*) (long) sig);
