@@ -44,6 +44,7 @@ def build_parser() -> argparse.ArgumentParser:
     )
     ap.add_argument("usr_command", nargs="?", default="", help="command to profile (record/stat)")
     ap.add_argument("--logdir", default="sofalog")
+    ap.add_argument("--version", action="version", version="sofa-amd 0.1.0")
     ap.add_argument("--verbose", action="store_true")
     # record
     ap.add_argument("--cpu_sample_rate", type=int, default=99)
