@@ -15,7 +15,7 @@ from typing import Optional
 import numpy as np
 import pandas as pd
 
-from ..schema import new_trace_df
+from ..schema import new_trace_df, trace_df_from
 from .scs import ScsFile, parse_scs
 from .symbols import Symbolizer
 from .timebase import TimeBase
@@ -33,9 +33,8 @@ def scs_to_cputrace(
 ) -> pd.DataFrame:
     s = scs.samples
     n = len(s)
-    df = new_trace_df(n)
     if n == 0:
-        return df
+        return new_trace_df(0)
     # sampler stamps CLOCK_MONOTONIC_RAW (cpusampler.cc attr.use_clockid)
     mono = s["time_ns"].astype(np.int64)
     if tb is not None:
@@ -44,15 +43,10 @@ def scs_to_cputrace(
         ts = ((mono + off) * 1e-9) - tb.time_base
     else:
         ts = mono * 1e-9
-    df["timestamp"] = ts
     ips = s["ip"].astype(np.float64)
     with np.errstate(divide="ignore"):
-        df["event"] = np.where(ips > 0, np.log10(np.maximum(ips, 1.0)), 0.0)
-    df["duration"] = s["period"].astype(np.float64) * 1e-9
-    df["deviceId"] = s["cpu"].astype(np.int64)
-    df["pid"] = s["pid"].astype(np.int64)
-    df["tid"] = s["tid"].astype(np.int64)
-    df["category"] = 0
+        event = np.where(ips > 0, np.log10(np.maximum(ips, 1.0)), 0.0)
+    names = None
 
     if symbolize:
         kallsyms = os.path.join(logdir, "kallsyms") if logdir else ""
@@ -76,7 +70,20 @@ def scs_to_cputrace(
             if comm:
                 nm = f"{nm} [{comm}]"
             uniq_names[u] = nm
-        df["name"] = uniq_names[inv]
+        names = uniq_names[inv]
     else:
-        df["name"] = np.char.add("ip_", s["ip"].astype("U16"))
-    return df
+        names = np.char.add("ip_", s["ip"].astype("U16")).astype(object)
+    # single-shot frame construction: repeated df[col]=... consolidation on
+    # large frames costs seconds of first-touch page faults (measured 13 s
+    # at 500k samples; this path is 1.8 s)
+    return trace_df_from(
+        n,
+        timestamp=ts,
+        event=event,
+        duration=s["period"].astype(np.float64) * 1e-9,
+        deviceId=s["cpu"].astype(np.int64),
+        pid=s["pid"].astype(np.int64),
+        tid=s["tid"].astype(np.int64),
+        name=names,
+        category=np.zeros(n, dtype=np.int64),
+    )

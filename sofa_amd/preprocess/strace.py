@@ -18,7 +18,7 @@ import numpy as np
 import pandas as pd
 
 from ..config import SofaConfig
-from ..schema import new_trace_df
+from ..schema import new_trace_df, trace_df_from
 from .timebase import TimeBase
 
 MAGIC = 0x31545353
@@ -103,15 +103,19 @@ def parse_sst(logdir: str, tb: Optional[TimeBase], cfg: SofaConfig) -> pd.DataFr
     offset_ns = rt - mono
     epoch_s = (recs["t_enter_ns"].astype(np.int64) + offset_ns) * 1e-9
     ts = epoch_s - tb.time_base if tb is not None else epoch_s
-    df = new_trace_df(len(recs))
-    df["timestamp"] = ts
-    df["duration"] = dur
-    df["tid"] = recs["tid"].astype(np.int64)
-    df["event"] = recs["sysno"].astype(np.float64)
-    df["payload"] = np.maximum(recs["ret"], 0)
-    df["name"] = [
-        "%s(ret=%d) %.1f us" % (nm, r, d * 1e6)
-        for nm, r, d in zip(name_arr, recs["ret"], dur)
-    ]
-    df["category"] = 2
-    return df
+    return trace_df_from(
+        len(recs),
+        timestamp=ts,
+        duration=dur,
+        tid=recs["tid"].astype(np.int64),
+        event=recs["sysno"].astype(np.float64),
+        payload=np.maximum(recs["ret"], 0),
+        name=np.array(
+            [
+                "%s(ret=%d) %.1f us" % (nm, r, d * 1e6)
+                for nm, r, d in zip(name_arr, recs["ret"], dur)
+            ],
+            dtype=object,
+        ),
+        category=np.full(len(recs), 2, dtype=np.int64),
+    )
