@@ -18,7 +18,7 @@ import numpy as np
 import pandas as pd
 
 from ..config import SofaConfig
-from ..schema import new_trace_df
+from ..schema import new_trace_df, trace_df_from
 from .timebase import TimeBase
 
 MAGIC = 0x31435053
@@ -75,9 +75,6 @@ def parse_pktcap(logdir: str, tb: Optional[TimeBase], cfg: SofaConfig) -> pd.Dat
     epoch_s = (pkts["time_ns"].astype(np.int64) + offset_ns) * 1e-9
     ts = epoch_s - tb.time_base if tb is not None else epoch_s
 
-    df = new_trace_df(n)
-    df["timestamp"] = ts
-    df["payload"] = pkts["len"].astype(np.int64)
     # duration model: payload / measured NIC bandwidth (fallback 1 GB/s)
     bw = 1e9
     nb_path = os.path.join(logdir, "netbandwidth.csv")
@@ -89,16 +86,24 @@ def parse_pktcap(logdir: str, tb: Optional[TimeBase], cfg: SofaConfig) -> pd.Dat
                 bw = float(peak)
         except (OSError, KeyError, ValueError):
             pass
-    df["duration"] = pkts["len"].astype(np.float64) / bw
-    df["bandwidth"] = bw
-    df["pkt_src"] = pack_ip_base1000(pkts["src_ip"])
-    df["pkt_dst"] = pack_ip_base1000(pkts["dst_ip"])
     proto_name = np.where(pkts["proto"] == 6, "tcp", np.where(pkts["proto"] == 17, "udp", "ip"))
-    df["name"] = [
-        "network:%s:%s:%d_to_%s:%d_with_%d" % (pr, ip_str(s), sp, ip_str(d), dp, ln)
-        for pr, s, sp, d, dp, ln in zip(
-            proto_name, pkts["src_ip"], pkts["sport"], pkts["dst_ip"], pkts["dport"], pkts["len"]
-        )
-    ]
-    df["category"] = 0
-    return df
+    names = np.array(
+        [
+            "network:%s:%s:%d_to_%s:%d_with_%d" % (pr, ip_str(s), sp, ip_str(d), dp, ln)
+            for pr, s, sp, d, dp, ln in zip(
+                proto_name, pkts["src_ip"], pkts["sport"], pkts["dst_ip"], pkts["dport"], pkts["len"]
+            )
+        ],
+        dtype=object,
+    )
+    return trace_df_from(
+        n,
+        timestamp=ts,
+        payload=pkts["len"].astype(np.int64),
+        duration=pkts["len"].astype(np.float64) / bw,
+        bandwidth=np.full(n, bw),
+        pkt_src=pack_ip_base1000(pkts["src_ip"]),
+        pkt_dst=pack_ip_base1000(pkts["dst_ip"]),
+        name=names,
+        category=np.zeros(n, dtype=np.int64),
+    )
