@@ -32,7 +32,9 @@ def main():
     ]
 
     print("n_records,produce_s,kernel_ms,records_per_s,effective_GBps,e2e_s,kept")
-    for log2n in (20, 22, 24, 25):
+    import os
+    sizes = (20, 22, 24, 25, 27) if os.environ.get("RING_BENCH_BIG") else (20, 22, 24, 25)
+    for log2n in sizes:
         n = 1 << log2n
         ring = ctypes.c_void_p()
         assert lib.sofa_ring_create(0, n, ctypes.byref(ring)) == 0
