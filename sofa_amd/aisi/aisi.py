@@ -138,7 +138,13 @@ def sofa_aisi(logdir, cfg, df_cpu, df_gpu, df_rccl, features,
     rows = []
     for i, s in enumerate(occ):
         t_begin = ts[s]
-        t_end = te[min(s + plen - 1, len(te) - 1)]
+        # an iteration spans from this occurrence to the next one — the mined
+        # pattern may cover only a stable subsequence of the step (autotuned
+        # kernels vary run to run), but occurrence STARTS delimit full steps
+        if i + 1 < len(occ):
+            t_end = ts[occ[i + 1]]
+        else:
+            t_end = te[min(s + plen - 1, len(te) - 1)]
         rows.append(iter_profile(df_gpu, df_rccl, t_begin, t_end))
     idf = pd.DataFrame(rows)
 

@@ -122,3 +122,35 @@ def test_aisi_via_strace(tmp_path):
     idf = sofa_aisi(str(tmp_path), cfg, None, None, None, feats, df_strace=df)
     assert idf is not None
     assert abs(len(idf) - n_iters) <= 2
+
+
+def test_aisi_tolerates_noise_kernels(tmp_path):
+    """10% random autotuning-style kernels inserted between iterations must
+    not break detection, and step boundaries = occurrence starts."""
+    rng = np.random.default_rng(3)
+    block = ["fwd_a", "gemm_b", "bwd_c", "opt_d"]
+    names, ts_list = [], []
+    t = 0.0
+    n_iters = 12
+    for it in range(n_iters):
+        for nm in block:
+            names.append(f"[gpu0] {nm}")
+            ts_list.append(t)
+            t += 1e-3
+        for _ in range(rng.integers(0, 2)):  # occasional stray kernel
+            names.append(f"[gpu0] autotune_{rng.integers(0, 1000)}")
+            ts_list.append(t)
+            t += 1e-3
+    df = new_trace_df(len(names))
+    df["timestamp"] = ts_list
+    df["duration"] = 9e-4
+    df["deviceId"] = 0
+    df["copyKind"] = 0
+    df["name"] = names
+    cfg = SofaConfig(logdir=str(tmp_path), num_iterations=n_iters)
+    feats = []
+    idf = sofa_aisi(str(tmp_path), cfg, None, df, None, feats)
+    assert idf is not None
+    assert abs(len(idf) - n_iters) <= 2
+    # step time ~ 4-5 ms (4 kernels + 0-1 stray)
+    assert 3.5e-3 < idf["step_time"].median() < 6e-3
