@@ -130,3 +130,24 @@ def test_demo_logdir_analyzes(tmp_path):
     shutil.copytree(demo, work)
     r = run_sofa(["analyze", "--logdir", work, "--skip_preprocess", "--no_gpu"])
     assert "Complete!!" in r.stdout
+
+
+def test_config_file_and_plugin(tmp_path, native_built):
+    """YAML --config + --plugins load path through the real CLI."""
+    cfgfile = tmp_path / "c.yaml"
+    cfgfile.write_text("sys_mon_rate: 25\nnum_swarms: 4\n")
+    plugdir = tmp_path / "plugs"
+    plugdir.mkdir()
+    (plugdir / "myplug.py").write_text(
+        "def myplug(cfg):\n    print('PLUGIN-SAW', cfg.num_swarms)\n"
+    )
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(plugdir) + ":" + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, SOFA, "stat", "sleep 0.4", "--logdir", str(tmp_path / "log"),
+         "--no_gpu", "--config", str(cfgfile), "--plugins", "myplug"],
+        capture_output=True, text=True, timeout=120, env=env,
+    )
+    assert "Complete!!" in r.stdout, r.stderr
+    # plugins run AFTER --config is applied, so they see the YAML overrides
+    assert "PLUGIN-SAW 4" in r.stdout
