@@ -93,3 +93,14 @@ def test_filter_parsing():
     fs = parse_filters("gemm:red,rccl:blue")
     assert fs[0].keyword == "gemm" and fs[0].color == "red"
     assert fs[1].keyword == "rccl" and fs[1].color == "blue"
+
+
+def test_trace_df_from_defaults_and_override():
+    from sofa_amd.schema import trace_df_from
+
+    df = trace_df_from(3, timestamp=np.array([1.0, 2.0, 3.0]), name=np.array(["a", "b", "c"], dtype=object))
+    assert list(df.columns) == TRACE_COLUMNS
+    assert df["deviceId"].tolist() == [-1, -1, -1]
+    assert df["copyKind"].tolist() == [-1, -1, -1]
+    assert df["name"].tolist() == ["a", "b", "c"]
+    assert df["timestamp"].tolist() == [1.0, 2.0, 3.0]
