@@ -53,7 +53,8 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--attach", type=int, default=0, metavar="PID",
                     help="record an already-running process instead of launching one")
     ap.add_argument("--duration", type=float, default=0.0,
-                    help="with --attach: stop after N seconds")
+                    help="stop recording after N seconds (works for launched "
+                    "commands and --attach)")
     ap.add_argument("--enable_tcpdump", action="store_true")
     ap.add_argument("--enable_strace", action="store_true")
     ap.add_argument("--enable_py_stacks", action="store_true")
@@ -209,7 +210,7 @@ def main(argv=None) -> int:
             p.print_error("record/stat needs a command to profile (or --attach PID)")
             return 2
         else:
-            sofa_record(args.usr_command, cfg)
+            sofa_record(args.usr_command, cfg, duration=args.duration)
         if verb == "record":
             return 0
 

@@ -242,7 +242,7 @@ def sofa_attach(pid: int, cfg: SofaConfig, duration: float = 0.0) -> int:
     return ret
 
 
-def sofa_record(command: str, cfg: SofaConfig) -> int:
+def sofa_record(command: str, cfg: SofaConfig, duration: float = 0.0) -> int:
     logdir = ensure_logdir(cfg)
     ensure_native_built(cfg.verbose)
     sofa_clean(cfg)
@@ -340,9 +340,22 @@ def sofa_record(command: str, cfg: SofaConfig) -> int:
         if pktcap_proc is not None:
             f.write("%d\n" % pktcap_proc.pid)
 
-    # --- wait ---
+    # --- wait (optionally time-boxed: a first-class version of the
+    # reference's EDR `sofa record "sleep N"` trick) ---
     try:
-        ret = target.wait()
+        if duration > 0:
+            try:
+                ret = target.wait(timeout=duration)
+            except subprocess.TimeoutExpired:
+                p.print_progress(f"duration {duration}s reached; stopping target")
+                target.terminate()
+                try:
+                    ret = target.wait(timeout=10)
+                except subprocess.TimeoutExpired:
+                    target.kill()
+                    ret = target.wait()
+        else:
+            ret = target.wait()
     except KeyboardInterrupt:
         target.send_signal(signal.SIGINT)
         ret = target.wait()

@@ -151,3 +151,15 @@ def test_config_file_and_plugin(tmp_path, native_built):
     assert "Complete!!" in r.stdout, r.stderr
     # plugins run AFTER --config is applied, so they see the YAML overrides
     assert "PLUGIN-SAW 4" in r.stdout
+
+
+def test_duration_timebox(tmp_path, native_built):
+    """--duration stops a long-running launched target."""
+    import time as _time
+
+    logdir = str(tmp_path / "log")
+    t0 = _time.time()
+    r = run_sofa(["stat", "sleep 30", "--logdir", logdir, "--no_gpu", "--duration", "1.5"])
+    took = _time.time() - t0
+    assert "Complete!!" in r.stdout, r.stderr
+    assert took < 25, f"duration not enforced ({took:.1f}s)"
