@@ -73,6 +73,19 @@ DERIVED_FILES = [
     "iteration_timeline.txt",
     "features.csv",
     "xlink_traffic.csv",
+    "kfdtrace.csv",
+    "markers.csv",
+    "strace.csv",
+    "pystacks.csv",
+    "chrome_trace.json",
+    "flamegraph.folded",
+    "gpu_timebase.json",
+    "comm_payload_matrix.csv",
+    "comm_bandwidth_matrix.csv",
+    "correlation.csv",
+    "cluster_report.csv",
+    "potato_report.html",
+    "swarm_diff.csv",
 ]
 
 
