@@ -40,7 +40,7 @@ def build_parser() -> argparse.ArgumentParser:
     )
     ap.add_argument(
         "command_verb",
-        choices=["stat", "record", "preprocess", "analyze", "report", "viz", "diff", "clean"],
+        choices=["stat", "record", "preprocess", "analyze", "report", "viz", "diff", "clean", "top"],
     )
     ap.add_argument("usr_command", nargs="?", default="", help="command to profile (record/stat)")
     ap.add_argument("--logdir", default="sofalog")
@@ -102,6 +102,8 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--match_logdir", default="sofalog-match")
     # viz
     ap.add_argument("--viz_port", type=int, default=8000)
+    ap.add_argument("--interval", type=float, default=1.0, help="top refresh seconds")
+    ap.add_argument("--once", action="store_true", help="top: one refresh then exit")
     ap.add_argument("--with-gui", dest="with_gui", action="store_true")
     # plugins (reference bin/sofa:21,322)
     ap.add_argument("--plugins", default="", help="comma-separated module names; each must expose f(cfg)")
@@ -194,6 +196,11 @@ def main(argv=None) -> int:
             return 2
     run_plugins(args, cfg)
     verb = args.command_verb
+
+    if verb == "top":
+        from .viz.top import sofa_top
+
+        return sofa_top(interval=args.interval, once=args.once)
 
     from .record import sofa_clean, sofa_record
 

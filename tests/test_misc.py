@@ -30,3 +30,15 @@ def test_timebase_loader_missing_dir(tmp_path):
     from sofa_amd.preprocess.timebase import load_timebase
 
     assert load_timebase(str(tmp_path)) is None  # no sofa_time.txt
+
+
+def test_sofa_top_once(capsys):
+    import time
+
+    from sofa_amd.viz.top import snapshot, sofa_top
+
+    prev, lines = snapshot({})
+    time.sleep(0.3)
+    prev, lines = snapshot(prev)
+    assert any("CPU" in ln for ln in lines)
+    assert any("NIC" in ln for ln in lines)
