@@ -61,3 +61,31 @@ def test_etl_bench_small():
     )
     assert r.returncode == 0, r.stderr[-1500:]
     assert "end-to-end ETL" in r.stdout
+
+
+def test_sofa_edr_triggers(tmp_path, monkeypatch):
+    """Event-driven recording: phase keyword in the watched log triggers a
+    time-boxed `sofa record` (subprocess stubbed)."""
+    edr = _load("sofa_edr", os.path.join(REPO, "tools", "sofa-edr.py"))
+    log = tmp_path / "app.log"
+    log.write_text("starting up\nphase: forward pass begins\n")
+    calls = []
+
+    def fake_run(cmd, check=False):
+        calls.append(cmd)
+
+        class R:
+            returncode = 0
+
+        return R()
+
+    monkeypatch.setattr(edr.subprocess, "run", fake_run)
+    monkeypatch.setattr(
+        edr.sys, "argv",
+        ["sofa-edr", "--watch", str(log), "--phases", "forward", "--duration", "5",
+         "--logdir-base", str(tmp_path / "edr"), "--poll", "0.05"],
+    )
+    edr.main()
+    assert len(calls) == 1
+    assert "sleep 5" in " ".join(calls[0])
+    assert any(str(tmp_path / "edr") + "-forward" in c for c in calls[0])
