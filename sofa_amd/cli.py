@@ -48,6 +48,8 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--verbose", action="store_true")
     # record
     ap.add_argument("--cpu_sample_rate", type=int, default=99)
+    ap.add_argument("--perf_events", default="cpu-clock",
+                    help="sampling event: cpu-clock (default) or cycles (needs a PMU)")
     ap.add_argument("--sys_mon_rate", type=int, default=10)
     ap.add_argument("--profile_all_cpus", action="store_true")
     ap.add_argument("--attach", type=int, default=0, metavar="PID",
@@ -133,6 +135,7 @@ def cfg_from_args(args) -> SofaConfig:
         command=args.usr_command,
         verbose=args.verbose,
         cpu_sample_rate=args.cpu_sample_rate,
+        perf_events=args.perf_events,
         sys_mon_rate=args.sys_mon_rate,
         profile_all_cpus=args.profile_all_cpus,
         enable_tcpdump=args.enable_tcpdump,

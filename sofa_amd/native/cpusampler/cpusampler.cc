@@ -333,6 +333,7 @@ int main(int argc, char** argv) {
   int freq = 99;
   pid_t target_pid = -1;  // -1 => system-wide
   bool system_wide = false;
+  bool want_hw_cycles = false;  // -e cycles: try the PMU, fall back to sw
   uint64_t max_mb = 512;
   for (int i = 1; i < argc; i++) {
     std::string a = argv[i];
@@ -342,6 +343,9 @@ int main(int argc, char** argv) {
     else if (a == "-a") system_wide = true;
     else if (a == "--max-mb" && i + 1 < argc) max_mb = strtoull(argv[++i], nullptr, 10);
     else if (a == "-g" || a == "--callchain") g_callchain = true;
+    else if (a == "-e" && i + 1 < argc) {
+      want_hw_cycles = (std::string(argv[++i]) == "cycles");
+    }
     else {
       fprintf(stderr, "usage: %s -o out.scs [-F hz] (-a | -p pid) [--max-mb N]\n", argv[0]);
       return 2;
@@ -380,6 +384,23 @@ int main(int argc, char** argv) {
   attr.type = PERF_TYPE_SOFTWARE;
   attr.size = sizeof(attr);
   attr.config = PERF_COUNT_SW_CPU_CLOCK;
+  if (want_hw_cycles) {
+    // probe the PMU once; VMs commonly expose none -> keep sw cpu-clock
+    struct perf_event_attr probe;
+    memset(&probe, 0, sizeof(probe));
+    probe.type = PERF_TYPE_HARDWARE;
+    probe.size = sizeof(probe);
+    probe.config = PERF_COUNT_HW_CPU_CYCLES;
+    probe.disabled = 1;
+    int pfd = (int) perf_event_open(&probe, 0, -1, -1, PERF_FLAG_FD_CLOEXEC);
+    if (pfd >= 0) {
+      close(pfd);
+      attr.type = PERF_TYPE_HARDWARE;
+      attr.config = PERF_COUNT_HW_CPU_CYCLES;
+    } else {
+      fprintf(stderr, "sofa-cpusampler: no hardware PMU; using cpu-clock\n");
+    }
+  }
   attr.freq = 1;
   attr.sample_freq = static_cast<uint64_t>(freq);
   attr.sample_type =

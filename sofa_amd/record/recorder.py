@@ -329,6 +329,8 @@ def sofa_record(command: str, cfg: SofaConfig, duration: float = 0.0) -> int:
     sampler_bin = native_bin("sofa-cpusampler")
     if os.path.exists(sampler_bin):
         args = [sampler_bin, "-o", os.path.join(logdir, "cpusamples.scs"), "-F", str(cfg.cpu_sample_rate)]
+        if cfg.perf_events and cfg.perf_events != "cpu-clock":
+            args += ["-e", cfg.perf_events]
         if cfg.enable_callchain:
             args.append("-g")
         if cfg.profile_all_cpus or cfg.enable_strace:
