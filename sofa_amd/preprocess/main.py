@@ -56,6 +56,7 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
         p.print_error(f"logdir {logdir} does not exist — run `sofa record` first")
         return {}
     p.print_progress(f"preprocessing {logdir}")
+    t_start = __import__("time").perf_counter()
     tb = load_timebase(logdir, cfg.cpu_time_offset_ms)
     traces: List[SOFATrace] = []
     result = {"tb": tb}
@@ -234,6 +235,10 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
     except Exception as e:
         p.print_warning(f"chrome trace export failed: {e}")
     n_events = sum(len(t.data) for t in traces if t.data is not None)
-    p.print_progress(f"preprocess done: {len(traces)} series, {n_events} points")
+    dt = __import__("time").perf_counter() - t_start
+    p.print_progress(
+        "preprocess done: %d series, %d points in %.2f s (%.0f events/s)"
+        % (len(traces), n_events, dt, n_events / max(dt, 1e-9))
+    )
     result["traces"] = traces
     return result
