@@ -154,3 +154,30 @@ def test_aisi_tolerates_noise_kernels(tmp_path):
     assert abs(len(idf) - n_iters) <= 2
     # step time ~ 4-5 ms (4 kernels + 0-1 stray)
     assert 3.5e-3 < idf["step_time"].median() < 6e-3
+
+
+def test_suffix_automaton_counts_property():
+    """Property: for random token strings, every candidate reported with
+    count k must actually occur (overlapping count) >= its non-overlapping
+    occurrence count, and exact patterns found via find_repeat_pattern truly
+    occur that many times."""
+    import random
+
+    from sofa_amd.aisi.stree import find_repeat_pattern, occurrences
+
+    rng = random.Random(7)
+    for _ in range(25):
+        n_sym = rng.randint(2, 5)
+        tokens = [rng.randrange(n_sym) for _ in range(rng.randint(20, 120))]
+        for k in (2, 3, 5):
+            for (start, length, cnt) in find_repeat_pattern(tokens, k, tol=0, min_len=2)[:5]:
+                pat = tokens[start : start + length]
+                # the automaton's endpos count = number of (possibly
+                # overlapping) occurrences; verify by brute force
+                brute = sum(
+                    1
+                    for i in range(len(tokens) - length + 1)
+                    if tokens[i : i + length] == pat
+                )
+                assert brute == cnt, (tokens, pat, cnt, brute)
+                assert len(occurrences(tokens, pat)) >= 1
