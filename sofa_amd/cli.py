@@ -104,6 +104,8 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--match_logdir", default="sofalog-match")
     # viz
     ap.add_argument("--viz_port", type=int, default=8000)
+    ap.add_argument("--viz_host", default="127.0.0.1",
+                    help="viz bind address (default loopback; set 0.0.0.0 to expose)")
     ap.add_argument("--interval", type=float, default=1.0, help="top refresh seconds")
     ap.add_argument("--once", action="store_true", help="top: one refresh then exit")
     ap.add_argument("--with-gui", dest="with_gui", action="store_true")
@@ -167,6 +169,7 @@ def cfg_from_args(args) -> SofaConfig:
         base_logdir=args.base_logdir,
         match_logdir=args.match_logdir,
         viz_port=args.viz_port,
+        viz_host=args.viz_host,
     )
     if args.gpu_filters:
         cfg.gpu_filters = parse_filters(args.gpu_filters)

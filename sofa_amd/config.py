@@ -121,6 +121,7 @@ class SofaConfig:
 
     # --- viz ---
     viz_port: int = 8000
+    viz_host: str = "127.0.0.1"
 
     # populated at runtime
     time_base: float = 0.0

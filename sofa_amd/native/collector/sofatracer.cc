@@ -85,9 +85,15 @@ void write_raw(const void* p, size_t n) {
   if (g_out) fwrite(p, 1, n, g_out);
 }
 
+// RecHeader.size is u16: clamp names so `total` cannot wrap the length
+// prefix and misalign every subsequent record (round-1 ADVICE).
+constexpr size_t kMaxNameLen = 65000;
+
 void write_name_rec(uint16_t type, uint64_t id, const char* name) {
   size_t len = name ? strlen(name) : 0;
+  if (len > kMaxNameLen) len = kMaxNameLen;
   size_t total = (sizeof(sgt::NameRec) + len + 1 + 7) & ~size_t(7);
+  static_assert(sizeof(sgt::NameRec) + kMaxNameLen + 8 <= UINT16_MAX, "");
   std::vector<char> buf(total, 0);
   auto* rec = reinterpret_cast<sgt::NameRec*>(buf.data());
   rec->h = {type, static_cast<uint16_t>(total), 0};
@@ -98,6 +104,7 @@ void write_name_rec(uint16_t type, uint64_t id, const char* name) {
 
 void write_opname_rec(uint32_t kind, uint32_t op, const char* name) {
   size_t len = name ? strlen(name) : 0;
+  if (len > kMaxNameLen) len = kMaxNameLen;
   size_t total = (sizeof(sgt::OpNameRec) + len + 1 + 7) & ~size_t(7);
   std::vector<char> buf(total, 0);
   auto* rec = reinterpret_cast<sgt::OpNameRec*>(buf.data());

@@ -367,18 +367,6 @@ int main(int argc, char** argv) {
     return 1;
   }
 
-  FileHeader hdr{};
-  hdr.magic = kMagic;
-  hdr.version = 1;
-  hdr.realtime_ns = clock_ns(CLOCK_REALTIME);
-  hdr.monotonic_raw_ns = clock_ns(CLOCK_MONOTONIC_RAW);
-  hdr.boottime_ns = clock_ns(CLOCK_BOOTTIME);
-  hdr.sample_freq = static_cast<uint32_t>(freq);
-  hdr.n_cpus = static_cast<uint32_t>(n_cpus);
-  fwrite(&hdr, sizeof(hdr), 1, f);
-
-  Writer writer(f, max_mb * (1ull << 20));
-
   struct perf_event_attr attr;
   memset(&attr, 0, sizeof(attr));
   attr.type = PERF_TYPE_SOFTWARE;
@@ -401,6 +389,23 @@ int main(int argc, char** argv) {
       fprintf(stderr, "sofa-cpusampler: no hardware PMU; using cpu-clock\n");
     }
   }
+
+  FileHeader hdr{};
+  hdr.magic = kMagic;
+  hdr.version = 1;
+  hdr.realtime_ns = clock_ns(CLOCK_REALTIME);
+  hdr.monotonic_raw_ns = clock_ns(CLOCK_MONOTONIC_RAW);
+  hdr.boottime_ns = clock_ns(CLOCK_BOOTTIME);
+  hdr.sample_freq = static_cast<uint32_t>(freq);
+  hdr.n_cpus = static_cast<uint32_t>(n_cpus);
+  // reserved[0] = sampled event unit so preprocess can convert periods:
+  // 0 = sw cpu-clock (period in ns), 1 = hw cycles (period in cycles;
+  // divide by MHz from cpuinfo.txt).  Round-1 ADVICE: without this, cycles
+  // periods were read as ns and inflated CPU busy time by the clock freq.
+  hdr.reserved[0] = (attr.type == PERF_TYPE_HARDWARE) ? 1 : 0;
+  fwrite(&hdr, sizeof(hdr), 1, f);
+
+  Writer writer(f, max_mb * (1ull << 20));
   attr.freq = 1;
   attr.sample_freq = static_cast<uint64_t>(freq);
   attr.sample_type =

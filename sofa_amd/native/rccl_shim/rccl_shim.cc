@@ -49,6 +49,19 @@ uint32_t elem_size(int dt) {
   return dt >= 0 && dt < (int) (sizeof(sz) / sizeof(sz[0])) ? sz[dt] : 0;
 }
 
+// Current HIP device of the calling thread, without linking HIP: librccl (or
+// the app) has already loaded libamdhip64, so resolve hipGetDevice lazily.
+// Every rank gets its own device id instead of collapsing onto GPU 0
+// (round-1 ADVICE: rccl_shim.cc device attribution).
+uint32_t current_device() {
+  using hipGetDevice_t = int (*)(int*);
+  static hipGetDevice_t fn =
+      reinterpret_cast<hipGetDevice_t>(dlsym(RTLD_DEFAULT, "hipGetDevice"));
+  int dev = 0;
+  if (fn && fn(&dev) == 0 && dev >= 0) return (uint32_t) dev;
+  return 0;
+}
+
 void ensure_open() {
   if (g_out) return;
   pthread_mutex_lock(&g_mutex);
@@ -110,7 +123,7 @@ void emit(uint32_t op, uint64_t t0, uint64_t t1, size_t count, int dt,
   r.datatype = (uint32_t) dt;
   r.elem_size = elem_size(dt);
   r.peer_or_root = peer_or_root;
-  r.device = 0;
+  r.device = current_device();
   r.comm = (uint64_t) comm;
   r.stream = (uint64_t) stream;
   pthread_mutex_lock(&g_mutex);

@@ -28,6 +28,25 @@ def load_scs(logdir: str) -> Optional[ScsFile]:
     return parse_scs(path)
 
 
+def mean_cpu_mhz(logdir: str) -> float:
+    """Mean sampled CPU MHz from cpuinfo.txt (recorder poller format:
+    'ts mhz mhz ...' per line).  Used to convert hw-cycles sample periods to
+    seconds, mirroring the reference's cycles/MHz conversion
+    (bin/sofa_preprocess.py:131-139).  0.0 when unavailable."""
+    path = os.path.join(logdir, "cpuinfo.txt")
+    try:
+        total, n = 0.0, 0
+        with open(path) as f:
+            for line in f:
+                parts = line.split()
+                for v in parts[1:]:
+                    total += float(v)
+                    n += 1
+        return total / n if n else 0.0
+    except (OSError, ValueError):
+        return 0.0
+
+
 def scs_to_cputrace(
     scs: ScsFile, tb: Optional[TimeBase], logdir: str = "", symbolize: bool = True
 ) -> pd.DataFrame:
@@ -54,11 +73,14 @@ def scs_to_cputrace(
             kallsyms = ""
         symr = Symbolizer(scs.mmaps, kallsyms)
         # resolve UNIQUE (pid, ip, kernel-flag) triples only, then map back
-        # vectorized (a long run has millions of samples but few unique IPs)
+        # vectorized (a long run has millions of samples but few unique IPs).
+        # Structured key — no bit-packing, so large pids (pid_max can be
+        # 4194304) cannot collide (round-1 ADVICE).
         kern = (s["flags"].astype(np.uint64) & 1)
-        key = (s["pid"].astype(np.uint64) << 49) | (kern << 48) | (
-            s["ip"].astype(np.uint64) & ((1 << 48) - 1)
-        )
+        key = np.empty(n, dtype=[("pid", "<u4"), ("kern", "<u1"), ("ip", "<u8")])
+        key["pid"] = s["pid"]
+        key["kern"] = kern
+        key["ip"] = s["ip"]
         # np.unique returns (values, first_indices, inverse) in this order
         uniq, first_idx, inv = np.unique(key, return_index=True, return_inverse=True)
         uniq_names = np.empty(len(uniq), dtype=object)
@@ -76,11 +98,20 @@ def scs_to_cputrace(
     # single-shot frame construction: repeated df[col]=... consolidation on
     # large frames costs seconds of first-touch page faults (measured 13 s
     # at 500k samples; this path is 1.8 s)
+    # period unit depends on the sampled event (scs header reserved[0]):
+    # cpu-clock periods are ns; hw-cycles periods are cycle counts and need
+    # the recorded MHz (reference converts cycles/MHz the same way)
+    if getattr(scs, "event_type", 0) == 1:
+        mhz = mean_cpu_mhz(logdir) if logdir else 0.0
+        hz = mhz * 1e6 if mhz > 0 else 2.0e9  # conservative 2 GHz fallback
+        duration = s["period"].astype(np.float64) / hz
+    else:
+        duration = s["period"].astype(np.float64) * 1e-9
     return trace_df_from(
         n,
         timestamp=ts,
         event=event,
-        duration=s["period"].astype(np.float64) * 1e-9,
+        duration=duration,
         deviceId=s["cpu"].astype(np.int64),
         pid=s["pid"].astype(np.int64),
         tid=s["tid"].astype(np.int64),

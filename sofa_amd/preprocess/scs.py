@@ -78,6 +78,7 @@ class ScsFile:
     boottime_ns: int = 0
     sample_freq: int = 0
     n_cpus: int = 0
+    event_type: int = 0  # header reserved[0]: 0 = cpu-clock (ns), 1 = cycles
     samples: np.ndarray = field(default_factory=lambda: np.empty(0, dtype=SAMPLE_DTYPE))
     samples_cs: np.ndarray = field(default_factory=lambda: np.empty(0, dtype=SAMPLE_CS_DTYPE))
     # pid -> list of (addr, len, pgoff, filename)
@@ -100,6 +101,9 @@ def parse_scs(path: str) -> ScsFile:
     out.boottime_ns = boot
     out.sample_freq = freq
     out.n_cpus = ncpus
+    # reserved[0] (u8 little-endian right after the packed header fields)
+    if len(buf) >= struct.calcsize(HEADER_FMT) + 8:
+        out.event_type = struct.unpack_from("<Q", buf, struct.calcsize(HEADER_FMT))[0]
 
     sample_chunks: List[np.ndarray] = []
     cs_chunks: List[np.ndarray] = []
