@@ -11,6 +11,77 @@ def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires an AMD GPU (MI355X) and ROCm runtime")
 
 
+def _warm_gpu():
+    """One-time GPU warmup on a fresh box, BEFORE any test's timeout starts.
+
+    Round-1 failure mode: the very first traced subprocess paid the cold
+    torch import + MIOpen/hipBLASLt kernel compilation (~3 min) and blew its
+    subprocess timeout, and -x fail-fast killed the whole suite.  Warming
+    here populates the on-disk MIOpen/hipBLASLt caches and the page cache,
+    which child processes share, and is not charged to any test because
+    pytest-timeout only covers test setup/call/teardown."""
+    try:
+        import torch
+
+        if not torch.cuda.is_available():
+            return
+    except Exception:
+        return
+    import subprocess
+    import time
+
+    import torch.nn as nn
+
+    t0 = time.time()
+    dev = "cuda:0"
+    try:
+        # shapes used by the gpu tests (small convs @64x64) and by
+        # smoke()/bench (resnet50 @224, channels_last, bf16 autocast)
+        m = nn.Sequential(
+            nn.Conv2d(3, 64, 7, 2, 3), nn.BatchNorm2d(64), nn.ReLU(),
+            nn.Conv2d(64, 64, 3, 1, 1),
+        ).to(dev)
+        x = torch.randn(16, 3, 64, 64, device=dev)
+        for _ in range(2):
+            m(x).square().mean().backward()
+        try:
+            from sofa_amd.workloads.resnet import build_resnet50
+
+            r50 = build_resnet50(device=dev)
+            xr = torch.randn(8, 3, 224, 224, device=dev).to(
+                memory_format=torch.channels_last
+            )
+            with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+                loss = r50(xr).float().square().mean()
+            loss.backward()
+            del r50, xr, loss
+        except Exception:
+            pass
+        a = torch.randn(1024, 1024, device=dev)
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+            (a @ a).sum().item()
+        torch.cuda.synchronize()
+        del m, x, a
+        torch.cuda.empty_cache()
+    except Exception as e:  # warmup must never fail the suite
+        print(f"[conftest] GPU warmup error (ignored): {e}")
+    # warm the child-interpreter path the traced subprocess tests take
+    try:
+        subprocess.run(
+            [sys.executable, "-c", "import torch; torch.cuda.is_available()"],
+            timeout=600,
+            capture_output=True,
+        )
+    except Exception:
+        pass
+    print(f"[conftest] GPU warmup done in {time.time() - t0:.1f}s", flush=True)
+
+
+def pytest_collection_finish(session):
+    if any(item.get_closest_marker("gpu") for item in session.items):
+        _warm_gpu()
+
+
 @pytest.fixture(scope="session")
 def repo_root():
     return REPO

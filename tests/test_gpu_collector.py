@@ -24,7 +24,7 @@ def _run_traced(cmd, logdir, extra_env=None):
     env["SOFA_LOGDIR"] = logdir
     env.update(extra_env or {})
     os.makedirs(logdir, exist_ok=True)
-    return subprocess.run(cmd, env=env, capture_output=True, text=True, timeout=300)
+    return subprocess.run(cmd, env=env, capture_output=True, text=True, timeout=900)
 
 
 def test_collector_traces_torch_kernels(tmp_path):
@@ -74,7 +74,7 @@ def test_sofa_stat_bandwidth_workload(tmp_path):
         ],
         capture_output=True,
         text=True,
-        timeout=600,
+        timeout=900,
     )
     assert "Complete!!" in r.stdout, (r.stdout[-3000:], r.stderr[-2000:])
     assert os.path.isfile(os.path.join(logdir, "gputrace.csv"))
@@ -113,7 +113,7 @@ def test_sofa_stat_torch_resnet_like(tmp_path):
         [sys.executable, SOFA, "stat", f"{sys.executable} -c \"{code}\"", "--logdir", logdir],
         capture_output=True,
         text=True,
-        timeout=600,
+        timeout=900,
     )
     assert "Complete!!" in r.stdout, (r.stdout[-3000:], r.stderr[-2000:])
     import pandas as pd

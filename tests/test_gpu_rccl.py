@@ -34,7 +34,7 @@ def test_rccl_allreduce_traced(tmp_path):
     )
     r = subprocess.run(
         [sys.executable, SOFA, "stat", f"{sys.executable} {snippet}", "--logdir", logdir],
-        capture_output=True, text=True, timeout=600,
+        capture_output=True, text=True, timeout=900,
     )
     assert "Complete!!" in r.stdout, (r.stdout[-2000:], r.stderr[-2000:])
     rccl_csv = os.path.join(logdir, "rccltrace.csv")
