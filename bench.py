@@ -177,20 +177,23 @@ def main() -> int:
         lib.sofa_tracer_stop()
         barrier_sync()
 
+        # chunk sizes sum EXACTLY to the requested step count (round 1
+        # silently rounded 20 down to 18; the driver's consistency check
+        # rightly flagged it)
         chunks = 3 if args.steps >= 6 else 1
-        per = max(args.steps // chunks, 1)
+        per, rem = divmod(args.steps, chunks)
+        chunk_sizes = [per + 1] * rem + [per] * (chunks - rem)
         t_prof = 0.0
         n0 = lib.sofa_tracer_event_count()
-        done_plain = done_prof = 0
-        for _ in range(chunks):
-            t_plain += timed_phase(per)
-            done_plain += per
+        done_prof = 0
+        for sz in chunk_sizes:
+            t_plain += timed_phase(sz)
             lib.sofa_tracer_start()
-            t_prof += timed_phase(per)
+            t_prof += timed_phase(sz)
             lib.sofa_tracer_stop()
-            done_prof += per
+            done_prof += sz
         n_events = int(lib.sofa_tracer_event_count() - n0)
-        args.steps = done_prof  # actual timed steps per mode
+        assert done_prof == args.steps, (done_prof, args.steps)
         if sampler is not None:
             sampler.terminate()
             sampler.wait(timeout=5)
@@ -240,6 +243,9 @@ def main() -> int:
                 "ms_per_step_plain": round(t_plain / args.steps * 1e3, 3),
                 "images_per_sec_profiled": round(imgs_per_sec, 1),
                 "trace_events": n_events,
+                # honesty: which world sizes THIS invocation measured (the
+                # driver runs N=1,2,4,8 itself for the scaling curve)
+                "measured_n_gpus": world_size,
             },
         }
         print(json.dumps(result))
