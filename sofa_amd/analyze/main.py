@@ -120,7 +120,7 @@ def sofa_analyze(cfg: SofaConfig, pre: Optional[dict] = None) -> Dict[str, float
     profiles.gpu_profile(df_gpu, df_rccl, features)
     profiles.gpusmi_profile(df_gpusmi, features, logdir)
     comm_mod.comm_profile(logdir, df_gpu, features)
-    comm_mod.rccl_link_attribution(logdir, df_rccl, topo, features)
+    comm_mod.rccl_link_attribution(logdir, df_rccl, topo, features, df_gpu=df_gpu)
 
     # --- launch-latency / launch-bound analysis (corr-id join) ---
     try:
