@@ -33,19 +33,56 @@ REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
 TRACER = os.path.join(REPO, "sofa_amd", "native", "lib", "libsofatracer.so")
+LITE = os.path.join(REPO, "sofa_amd", "native", "lib", "libsofahsalite.so")
 
 
-def setup_tracer_env(logdir: str) -> None:
+def setup_tracer_env(logdir: str, tracer_mode: str) -> None:
     """Must run BEFORE importing torch (the HIP runtime registers tools at
     init)."""
     os.environ["SOFA_LOGDIR"] = logdir
     os.environ["SOFA_DEFER_START"] = "1"
-    os.environ["SOFA_TRACE_HIP_API"] = "1"
-    os.environ["SOFA_TRACE_RCCL"] = "1"
-    os.environ.setdefault("SOFA_GPU_BUFFER_MB", "64")
+    if tracer_mode == "lite":
+        # HSA-level dispatch/copy tracer + SDK collector demoted to
+        # RCCL-args only (sofa_amd/native/hsalite/hsalite.cc)
+        prev_hsa = os.environ.get("HSA_TOOLS_LIB", "")
+        if LITE not in prev_hsa:
+            os.environ["HSA_TOOLS_LIB"] = LITE + ((" " + prev_hsa) if prev_hsa else "")
+        os.environ["SOFA_TRACE_DISPATCH"] = "0"
+        os.environ["SOFA_TRACE_RCCL"] = "1"
+    else:
+        os.environ["SOFA_TRACE_HIP_API"] = "1"
+        os.environ["SOFA_TRACE_RCCL"] = "1"
+        os.environ.setdefault("SOFA_GPU_BUFFER_MB", "64")
     prev = os.environ.get("ROCP_TOOL_LIBRARIES", "")
     if TRACER not in prev:
         os.environ["ROCP_TOOL_LIBRARIES"] = TRACER + ((":" + prev) if prev else "")
+
+
+class TracerCtl:
+    """Unified start/stop/event-count over the active tracer libraries."""
+
+    def __init__(self, tracer_mode: str):
+        self.libs = []
+        lib = ctypes.CDLL(TRACER)
+        lib.sofa_tracer_event_count.restype = ctypes.c_ulonglong
+        self.libs.append(("sofa_tracer", lib))
+        if tracer_mode == "lite":
+            ll = ctypes.CDLL(LITE)
+            ll.sofa_lite_event_count.restype = ctypes.c_ulonglong
+            self.libs.append(("sofa_lite", ll))
+
+    def start(self):
+        for prefix, lib in self.libs:
+            getattr(lib, prefix + "_start")()
+
+    def stop(self):
+        for prefix, lib in self.libs:
+            getattr(lib, prefix + "_stop")()
+
+    def event_count(self):
+        return sum(
+            int(getattr(lib, prefix + "_event_count")()) for prefix, lib in self.libs
+        )
 
 
 def main() -> int:
@@ -55,6 +92,10 @@ def main() -> int:
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--no-profile", action="store_true", help="skip the profiled phase")
+    ap.add_argument("--tracer", choices=["sdk", "lite"],
+                    default=os.environ.get("SOFA_BENCH_TRACER", "lite"),
+                    help="profiled-phase collector: lite = HSA-level tracer "
+                    "(default), sdk = rocprofiler-sdk collector")
     ap.add_argument("--hip-api", type=int, default=1,
                     help="trace HIP runtime API spans (filtered op set; "
                     "SOFA_HIP_API_OPS=all for every call)")
@@ -71,10 +112,13 @@ def main() -> int:
     logdir = os.path.join(REPO, "gpurun_out", "bench_sgt")
     if rank == 0:
         os.makedirs(logdir, exist_ok=True)
+    if args.tracer == "lite" and not os.path.exists(LITE):
+        args.tracer = "sdk"
     have_tracer = os.path.exists(TRACER)
     if have_tracer:
-        setup_tracer_env(logdir)
-        os.environ["SOFA_TRACE_HIP_API"] = "1" if args.hip_api else "0"
+        setup_tracer_env(logdir, args.tracer)
+        if args.tracer == "sdk":
+            os.environ["SOFA_TRACE_HIP_API"] = "1" if args.hip_api else "0"
 
     import torch  # AFTER env setup
     import torch.distributed as dist
@@ -146,10 +190,8 @@ def main() -> int:
     t_plain = 0.0
     t_prof = None
     n_events = 0
-    lib = None
     if have_tracer and use_cuda and not args.no_profile:
-        lib = ctypes.CDLL(TRACER)
-        lib.sofa_tracer_event_count.restype = ctypes.c_ulonglong
+        lib = TracerCtl(args.tracer)
         import subprocess
 
         mon = None
@@ -171,10 +213,10 @@ def main() -> int:
                 )
         # profiled warmup: first traced launches pay one-time interception
         # setup; keep that out of the timed region
-        lib.sofa_tracer_start()
+        lib.start()
         for _ in range(max(2, args.warmup // 2)):
             step()
-        lib.sofa_tracer_stop()
+        lib.stop()
         barrier_sync()
 
         # chunk sizes sum EXACTLY to the requested step count (round 1
@@ -184,15 +226,15 @@ def main() -> int:
         per, rem = divmod(args.steps, chunks)
         chunk_sizes = [per + 1] * rem + [per] * (chunks - rem)
         t_prof = 0.0
-        n0 = lib.sofa_tracer_event_count()
+        n0 = lib.event_count()
         done_prof = 0
         for sz in chunk_sizes:
             t_plain += timed_phase(sz)
-            lib.sofa_tracer_start()
+            lib.start()
             t_prof += timed_phase(sz)
-            lib.sofa_tracer_stop()
+            lib.stop()
             done_prof += sz
-        n_events = int(lib.sofa_tracer_event_count() - n0)
+        n_events = int(lib.event_count() - n0)
         assert done_prof == args.steps, (done_prof, args.steps)
         if sampler is not None:
             sampler.terminate()

@@ -63,6 +63,13 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--call_stacks", action="store_true",
                     help="sample native call stacks (-g) -> flamegraph.folded + flame.html")
     ap.add_argument("--no_gpu", action="store_true", help="disable GPU tracing")
+    ap.add_argument(
+        "--gpu_tracer",
+        choices=["sdk", "lite"],
+        default="sdk",
+        help="sdk = rocprofiler-sdk collector (full fidelity); lite = "
+        "HSA-level dispatch tracer (lowest overhead) + SDK RCCL-args only",
+    )
     ap.add_argument("--no_hip_api", action="store_true",
                     help="disable HIP runtime API span tracing")
     ap.add_argument(
@@ -147,6 +154,7 @@ def cfg_from_args(args) -> SofaConfig:
         enable_gpu=not args.no_gpu,
         enable_gpu_hip_api=not args.no_hip_api,
         hip_api_full=args.hip_api_full,
+        gpu_tracer=args.gpu_tracer,
         enable_rccl_trace=not args.no_rccl,
         rccl_shim=args.rccl_shim,
         gpu_ring_buffer_mb=args.gpu_buffer_mb,

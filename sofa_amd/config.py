@@ -84,6 +84,11 @@ class SofaConfig:
     # copies/syncs/allocs — measured ~0 extra overhead at 2x events/s);
     # full API tracing (--hip_api_full) costs ~45% on launch-dense steps
     # (profiles/overhead_matrix_r01.md)
+    # "sdk" = rocprofiler-sdk collector (full fidelity: HIP API spans, KFD,
+    # allocs); "lite" = HSA-level dispatch tracer (lowest overhead) + SDK
+    # demoted to RCCL-args/markers only.  Flipped to lite-by-default once
+    # GPU-validated.
+    gpu_tracer: str = "sdk"
     enable_gpu_hip_api: bool = True
     hip_api_full: bool = False
     enable_rccl_trace: bool = True     # RCCL API tracing via collector

@@ -86,6 +86,26 @@ def build_all(verbose: bool = False) -> None:
             ],
             [f"-L{ROCM}/lib", "-lrocprofiler-sdk", f"-Wl,-rpath,{ROCM}/lib"],
         ),
+        (
+            os.path.join(LIB, "libsofahsalite.so"),
+            [
+                os.path.join(HERE, "hsalite", "hsalite.cc"),
+                os.path.join(HERE, "collector", "sgt_format.h"),
+            ],
+            [
+                "g++",
+                "-O2",
+                "-std=c++17",
+                "-fPIC",
+                "-shared",
+                # hsa_api_trace.h uses the in-tree "inc/" include layout
+                # unless AMD_INTERNAL_BUILD; the installed headers are flat
+                "-DAMD_INTERNAL_BUILD",
+                f"-I{ROCM}/include",
+                f"-I{ROCM}/include/hsa",
+            ],
+            ["-lpthread"],
+        ),
     ]
     for dst, srcs, cc, link in host_targets:
         if not os.path.exists(srcs[0]):

@@ -536,11 +536,18 @@ int tool_init(rocprofiler_client_finalize_t, void*) {
                                 &g_buffer) != ROCPROFILER_STATUS_SUCCESS)
     return -1;
 
-  rocprofiler_configure_buffer_tracing_service(
-      g_ctx, ROCPROFILER_BUFFER_TRACING_KERNEL_DISPATCH, nullptr, 0, g_buffer);
-  rocprofiler_configure_buffer_tracing_service(
-      g_ctx, ROCPROFILER_BUFFER_TRACING_MEMORY_COPY, nullptr, 0, g_buffer);
-  if (env_flag("SOFA_TRACE_HIP_API", true)) {
+  // SOFA_TRACE_DISPATCH=0: RCCL/marker-only mode — no kernel/copy buffer
+  // services, so the SDK installs no per-dispatch interception and this
+  // library composes with libsofahsalite (which owns the dispatch timeline
+  // at a fraction of the cost; see hsalite/hsalite.cc)
+  bool trace_dispatch = env_flag("SOFA_TRACE_DISPATCH", true);
+  if (trace_dispatch) {
+    rocprofiler_configure_buffer_tracing_service(
+        g_ctx, ROCPROFILER_BUFFER_TRACING_KERNEL_DISPATCH, nullptr, 0, g_buffer);
+    rocprofiler_configure_buffer_tracing_service(
+        g_ctx, ROCPROFILER_BUFFER_TRACING_MEMORY_COPY, nullptr, 0, g_buffer);
+  }
+  if (trace_dispatch && env_flag("SOFA_TRACE_HIP_API", true)) {
     // SOFA_HIP_API_OPS: "all" = every HIP call (expensive at launch rates),
     // default = launches/copies/syncs/allocs only (the calls a timeline
     // reader actually drills into; cuts record volume massively)

@@ -110,13 +110,19 @@ def sgt_to_gputrace(
                 resolved = {kid: demangle(nm) for kid, nm in names.items()}
             else:
                 resolved = names
-            # vectorized naming: format only unique (device, kernel_id) pairs
-            pairs = (k["device"].astype(np.int64) << 32) | k["kernel_id"].astype(np.int64)
-            uniq, inv = np.unique(pairs, return_inverse=True)
+            # vectorized naming: format only unique (device, kernel_id) pairs.
+            # Structured key, not bit-packing: the lite collector's kernel_id
+            # is the 64-bit HSA kernel_object handle, which does not fit in
+            # 32 bits.
+            key = np.empty(len(k), dtype=[("dev", "<i8"), ("kid", "<u8")])
+            key["dev"] = k["device"]
+            key["kid"] = k["kernel_id"]
+            uniq, inv = np.unique(key, return_inverse=True)
             uniq_names = np.array(
                 [
-                    "[gpu%d] %s" % (p >> 32, resolved.get(p & 0xFFFFFFFF, "kernel_%d" % (p & 0xFFFFFFFF)))
-                    for p in uniq
+                    "[gpu%d] %s"
+                    % (u["dev"], resolved.get(int(u["kid"]), "kernel_%d" % u["kid"]))
+                    for u in uniq
                 ],
                 dtype=object,
             )

@@ -165,7 +165,23 @@ def build_target_env(cfg: SofaConfig) -> dict:
     env = dict(os.environ)
     if cfg.enable_gpu:
         tracer = native_lib("libsofatracer.so")
-        if os.path.exists(tracer):
+        lite = native_lib("libsofahsalite.so")
+        mode = getattr(cfg, "gpu_tracer", "sdk")
+        if mode == "lite" and not os.path.exists(lite):
+            mode = "sdk"  # graceful fallback
+        if mode == "lite" and os.path.exists(lite):
+            # default: HSA-level dispatch/copy tracer (lowest overhead;
+            # hsalite/hsalite.cc) + the SDK collector demoted to
+            # RCCL-args/markers only (no per-dispatch interception)
+            prev_hsa = env.get("HSA_TOOLS_LIB", "")
+            env["HSA_TOOLS_LIB"] = lite + ((" " + prev_hsa) if prev_hsa else "")
+            env["SOFA_LOGDIR"] = os.path.abspath(cfg.logdir)
+            if os.path.exists(tracer):
+                prev = env.get("ROCP_TOOL_LIBRARIES", "")
+                env["ROCP_TOOL_LIBRARIES"] = tracer + ((":" + prev) if prev else "")
+                env["SOFA_TRACE_DISPATCH"] = "0"
+                env["SOFA_TRACE_RCCL"] = "1" if cfg.enable_rccl_trace else "0"
+        elif os.path.exists(tracer):
             prev = env.get("ROCP_TOOL_LIBRARIES", "")
             env["ROCP_TOOL_LIBRARIES"] = tracer + ((":" + prev) if prev else "")
             env["SOFA_LOGDIR"] = os.path.abspath(cfg.logdir)

@@ -1,6 +1,10 @@
 """Small coverage tests: CPU-degradation paths of GPU-facing helpers."""
 
+import os
+
 import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def test_gpu_timebase_none_on_cpu():
@@ -42,3 +46,18 @@ def test_sofa_top_once(capsys):
     prev, lines = snapshot(prev)
     assert any("CPU" in ln for ln in lines)
     assert any("NIC" in ln for ln in lines)
+
+
+def test_hsalite_library_loads():
+    """libsofahsalite must dlopen on any box (no GPU needed) and stay
+    passive until ROCr calls OnLoad."""
+    import ctypes
+
+    lib_path = os.path.join(
+        REPO, "sofa_amd", "native", "lib", "libsofahsalite.so"
+    )
+    assert os.path.exists(lib_path), "build_all did not produce libsofahsalite"
+    lib = ctypes.CDLL(lib_path)
+    assert lib.sofa_lite_active() == 0
+    lib.sofa_lite_event_count.restype = ctypes.c_ulonglong
+    assert lib.sofa_lite_event_count() == 0
