@@ -115,6 +115,10 @@ def sofa_analyze(cfg: SofaConfig, pre: Optional[dict] = None) -> Dict[str, float
     profiles.mpstat_profile(df_mpstat, features, cfg.is_idle_threshold)
     profiles.vmstat_profile(logdir, features)
     profiles.diskstat_profile(logdir, features)
+    df_blkio = pre.get("df_blkio")
+    if df_blkio is None or len(df_blkio) == 0:
+        df_blkio = _load_csv_trace(logdir, "blktrace.csv")
+    profiles.blkio_latency_profile(df_blkio, features)
     profiles.netbandwidth_profile(logdir, features)
     profiles.net_profile(logdir, df_net, features)
     profiles.gpu_profile(df_gpu, df_rccl, features)

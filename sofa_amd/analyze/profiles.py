@@ -176,6 +176,32 @@ def diskstat_profile(logdir: str, features: Features) -> None:
     features.append(("disk_write_Bps", float(d["write_Bps"].mean())))
 
 
+def blkio_latency_profile(df_blk: pd.DataFrame, features: Features) -> None:
+    """Per-IO latency summary per device (reference blktrace_latency_profile,
+    bin/sofa_analyze.py:596-638 — btt replaced by our matched issue/complete
+    rows from preprocess.blkio)."""
+    if df_blk is None or len(df_blk) == 0:
+        return
+    print("\nBlock-IO latency profile (per request):")
+    print("%-10s %8s %12s %10s %10s %10s %12s" % (
+        "device", "IOs", "bytes(MB)", "q25(ms)", "q50(ms)", "q95(ms)", "MB/s"))
+    for dev, grp in df_blk.groupby("deviceId"):
+        maj, minr = int(dev) >> 20, int(dev) & 0xFFFFF
+        lat = grp["duration"]
+        q25, q50, _, _ = _q(lat)
+        q95 = float(lat.quantile(0.95))
+        total_b = grp["payload"].sum()
+        total_t = lat.sum()
+        mbps = total_b / max(total_t, 1e-9) / 1e6
+        print("%-10s %8d %12.2f %10.3f %10.3f %10.3f %12.1f" % (
+            f"{maj},{minr}", len(grp), total_b / 1e6,
+            q25 * 1e3, q50 * 1e3, q95 * 1e3, mbps))
+    features.append(("blkio_num_requests", float(len(df_blk))))
+    features.append(("blkio_latency_q50", float(df_blk["duration"].quantile(0.5))))
+    features.append(("blkio_latency_q95", float(df_blk["duration"].quantile(0.95))))
+    features.append(("blkio_total_bytes", float(df_blk["payload"].sum())))
+
+
 def netbandwidth_profile(logdir: str, features: Features) -> None:
     path = os.path.join(logdir, "netbandwidth.csv")
     if not os.path.isfile(path):

@@ -86,6 +86,10 @@ def build_parser() -> argparse.ArgumentParser:
         "the rocprofiler-sdk path",
     )
     ap.add_argument("--gpu_buffer_mb", type=int, default=64)
+    ap.add_argument("--enable_blkio", action="store_true",
+                    help="per-request block-IO latency via tracefs "
+                    "block_rq_issue/complete (blktrace parity)")
+    ap.add_argument("--blkdev", default="", help="block device hint (implies --enable_blkio)")
     ap.add_argument("--enable_kfd_trace", action="store_true",
                     help="trace KFD page-migrate/fault events (SVM memory pressure)")
     # preprocess
@@ -155,6 +159,8 @@ def cfg_from_args(args) -> SofaConfig:
         enable_gpu_hip_api=not args.no_hip_api,
         hip_api_full=args.hip_api_full,
         gpu_tracer=args.gpu_tracer,
+        enable_blkio=args.enable_blkio,
+        blkdev=args.blkdev,
         enable_rccl_trace=not args.no_rccl,
         rccl_shim=args.rccl_shim,
         gpu_ring_buffer_mb=args.gpu_buffer_mb,

@@ -96,6 +96,7 @@ class SofaConfig:
     enable_kfd_trace: bool = False     # page-migrate/fault events
     gpu_ring_buffer_mb: int = 64       # collector buffer size per process
     blkdev: str = ""                   # block device for blktrace-like stats
+    enable_blkio: bool = False         # tracefs block_rq_issue/complete per-IO tracing
     nvsmi_interval_ms: int = 100       # GPU telemetry poll period
 
     # --- preprocess ---

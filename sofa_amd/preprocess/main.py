@@ -127,6 +127,21 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
         )
     result["df_diskstat"] = t_disk
 
+    # per-request block-IO latency (tracefs stream; blktrace parity)
+    try:
+        from .blkio import parse_blkio
+
+        df_blk = parse_blkio(logdir, tb)
+        if len(df_blk):
+            write_trace_csv(df_blk, os.path.join(logdir, "blktrace.csv"))
+            traces.append(
+                SOFATrace(name="blkio_traces", title="Block IO latency", color="Peru", data=df_blk)
+            )
+        result["df_blkio"] = df_blk
+    except Exception as e:
+        p.print_warning(f"blkio parse failed: {e}")
+        result["df_blkio"] = new_trace_df(0)
+
     t_net, net_bw = sysmon.parse_netstat(logdir, tb)
     if len(t_net):
         traces.append(SOFATrace(name="netstat_traces", title="NIC throughput (MB/s)", color="YellowGreen", data=t_net))

@@ -300,6 +300,13 @@ def sofa_record(command: str, cfg: SofaConfig, duration: float = 0.0) -> int:
     mon = SysMonitor(logdir, rate_hz=cfg.sys_mon_rate, enable_gpu=cfg.enable_gpu)
     mon.start()
 
+    blk = None
+    if getattr(cfg, "enable_blkio", False) or cfg.blkdev:
+        from .blkio import BlkTracer
+
+        blk = BlkTracer(logdir, device=cfg.blkdev)
+        blk.start()
+
     pktcap_proc = None
     if cfg.enable_tcpdump:
         sniffer = native_bin("sofa-pktcap")
@@ -405,6 +412,8 @@ def sofa_record(command: str, cfg: SofaConfig, duration: float = 0.0) -> int:
             pktcap_proc.wait(timeout=5)
         except (OSError, subprocess.TimeoutExpired):
             pktcap_proc.kill()
+    if blk is not None:
+        blk.stop()
     mon.stop()
     mon.join(timeout=5)
     err_f.close()
