@@ -48,6 +48,7 @@ def setup_tracer_env(logdir: str, tracer_mode: str) -> None:
         if LITE not in prev_hsa:
             os.environ["HSA_TOOLS_LIB"] = LITE + ((" " + prev_hsa) if prev_hsa else "")
         os.environ["SOFA_TRACE_DISPATCH"] = "0"
+        os.environ["SOFA_TRACE_COPY"] = "1"
         os.environ["SOFA_TRACE_RCCL"] = "1"
     else:
         os.environ["SOFA_TRACE_HIP_API"] = "1"

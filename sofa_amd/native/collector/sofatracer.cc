@@ -544,6 +544,12 @@ int tool_init(rocprofiler_client_finalize_t, void*) {
   if (trace_dispatch) {
     rocprofiler_configure_buffer_tracing_service(
         g_ctx, ROCPROFILER_BUFFER_TRACING_KERNEL_DISPATCH, nullptr, 0, g_buffer);
+  }
+  // Copies are low-rate (tens per step), so SDK copy tracing is ~free and
+  // stays on in the lite hybrid (SOFA_TRACE_COPY=1 SOFA_TRACE_DISPATCH=0):
+  // measured on MI355X, ROCclr's D2H/H2D SDMA path does not go through the
+  // public hsa_amd_memory_async_copy* entries hsalite wraps.
+  if (env_flag("SOFA_TRACE_COPY", trace_dispatch)) {
     rocprofiler_configure_buffer_tracing_service(
         g_ctx, ROCPROFILER_BUFFER_TRACING_MEMORY_COPY, nullptr, 0, g_buffer);
   }

@@ -233,6 +233,20 @@ constexpr uint32_t kSlotsPerQueue = 8192;
 
 std::atomic<uint64_t> g_pool_exhausted{0};
 
+// diagnostics (SOFA_LITE_DEBUG=1 prints at exit; sofa_lite_stats exports)
+struct Stats {
+  std::atomic<uint64_t> queue_create_gpu{0};
+  std::atomic<uint64_t> submit_batches{0};
+  std::atomic<uint64_t> kernel_pkts{0};
+  std::atomic<uint64_t> attached{0};
+  std::atomic<uint64_t> skipped_has_signal{0};
+  std::atomic<uint64_t> reaped{0};
+  std::atomic<uint64_t> copy_calls{0};
+  std::atomic<uint64_t> copy_engine_calls{0};
+  std::atomic<uint64_t> copy_rect_calls{0};
+  std::atomic<uint64_t> copies_recorded{0};
+} g_stats;
+
 pthread_t g_reaper;
 std::atomic<bool> g_reaper_started{false};
 
@@ -257,6 +271,7 @@ void on_submit(const void* pkts, uint64_t pkt_count, uint64_t /*user_pkt_index*/
     writer(pkts, pkt_count);
     return;
   }
+  g_stats.submit_batches.fetch_add(1, std::memory_order_relaxed);
   // find kernel dispatch packets we can instrument
   const auto* in = static_cast<const hsa_kernel_dispatch_packet_t*>(pkts);
   static thread_local std::vector<hsa_kernel_dispatch_packet_t> scratch;
@@ -264,8 +279,12 @@ void on_submit(const void* pkts, uint64_t pkt_count, uint64_t /*user_pkt_index*/
   for (uint64_t i = 0; i < pkt_count; ++i) {
     uint8_t type = packet_type(in[i].header);
     if (type != HSA_PACKET_TYPE_KERNEL_DISPATCH) continue;
+    g_stats.kernel_pkts.fetch_add(1, std::memory_order_relaxed);
     bool has_sig = in[i].completion_signal.handle != 0;
-    if (has_sig && !g_replace_signals) continue;
+    if (has_sig && !g_replace_signals) {
+      g_stats.skipped_has_signal.fetch_add(1, std::memory_order_relaxed);
+      continue;
+    }
 
     // acquire a slot + signal
     pthread_spin_lock(&q->lock);
@@ -300,6 +319,7 @@ void on_submit(const void* pkts, uint64_t pkt_count, uint64_t /*user_pkt_index*/
     pthread_spin_lock(&q->lock);
     q->inflight.push_back(idx);
     pthread_spin_unlock(&q->lock);
+    g_stats.attached.fetch_add(1, std::memory_order_relaxed);
   }
   writer(patched ? (const void*) scratch.data() : pkts, pkt_count);
 }
@@ -366,6 +386,7 @@ void* reaper_main(void*) {
           chunk.insert(chunk.end(), (const char*) &rec,
                        (const char*) &rec + sizeof(rec));
           ++n_reaped;
+          g_stats.reaped.fetch_add(1, std::memory_order_relaxed);
         }
         // recycle
         g_core.hsa_signal_store_screlease_fn(s.sig, 1);
@@ -405,6 +426,8 @@ hsa_status_t queue_create_wrap(hsa_agent_t agent, uint32_t size,
   enumerate_agents();
   hsa_device_type_t dev_type = HSA_DEVICE_TYPE_CPU;
   g_core.hsa_agent_get_info_fn(agent, HSA_AGENT_INFO_DEVICE, &dev_type);
+  if (dev_type == HSA_DEVICE_TYPE_GPU)
+    g_stats.queue_create_gpu.fetch_add(1, std::memory_order_relaxed);
   if (dev_type != HSA_DEVICE_TYPE_GPU || g_mode == MODE_OFF || !g_out) {
     // CPU soft queues / disabled: plain queue (MODE_OFF still proxies below
     // to measure the proxy floor — but with no handler registered)
@@ -569,6 +592,7 @@ bool copy_done_handler(hsa_signal_value_t, void* arg) {
     rec.bytes = s->bytes;
     write_raw(&rec, sizeof(rec));
     g_n_records.fetch_add(1, std::memory_order_relaxed);
+    g_stats.copies_recorded.fetch_add(1, std::memory_order_relaxed);
   }
   copy_slot_put(s);
   return false;  // one-shot
@@ -579,6 +603,7 @@ hsa_status_t async_copy_wrap(void* dst, hsa_agent_t dst_agent, const void* src,
                              uint32_t num_dep_signals,
                              const hsa_signal_t* dep_signals,
                              hsa_signal_t completion_signal) {
+  g_stats.copy_calls.fetch_add(1, std::memory_order_relaxed);
   if (g_mode != MODE_FULL || !g_armed.load(std::memory_order_relaxed) ||
       completion_signal.handle == 0) {
     return g_amd.hsa_amd_memory_async_copy_fn(dst, dst_agent, src, src_agent,
@@ -620,6 +645,7 @@ hsa_status_t async_copy_engine_wrap(void* dst, hsa_agent_t dst_agent,
                                     hsa_signal_t completion_signal,
                                     hsa_amd_sdma_engine_id_t engine_id,
                                     bool force_copy_on_sdma) {
+  g_stats.copy_engine_calls.fetch_add(1, std::memory_order_relaxed);
   if (g_mode != MODE_FULL || !g_armed.load(std::memory_order_relaxed) ||
       completion_signal.handle == 0) {
     return g_amd.hsa_amd_memory_async_copy_on_engine_fn(
@@ -650,6 +676,22 @@ hsa_status_t async_copy_engine_wrap(void* dst, hsa_agent_t dst_agent,
   g_amd.hsa_amd_signal_async_handler_fn(s->sig, HSA_SIGNAL_CONDITION_LT, 1,
                                         copy_done_handler, s);
   return st;
+}
+
+hsa_status_t async_copy_rect_wrap(const hsa_pitched_ptr_t* dst,
+                                  const hsa_dim3_t* dst_offset,
+                                  const hsa_pitched_ptr_t* src,
+                                  const hsa_dim3_t* src_offset,
+                                  const hsa_dim3_t* range,
+                                  hsa_agent_t copy_agent,
+                                  hsa_amd_copy_direction_t dir,
+                                  uint32_t num_dep_signals,
+                                  const hsa_signal_t* dep_signals,
+                                  hsa_signal_t completion_signal) {
+  g_stats.copy_rect_calls.fetch_add(1, std::memory_order_relaxed);
+  return g_amd.hsa_amd_memory_async_copy_rect_fn(
+      dst, dst_offset, src, src_offset, range, copy_agent, dir,
+      num_dep_signals, dep_signals, completion_signal);
 }
 
 // ------------------------------------------------------------------ control
@@ -685,6 +727,18 @@ void finalize() {
   if (!done.compare_exchange_strong(expected, true)) return;
   g_shutdown.store(true, std::memory_order_release);
   if (g_reaper_started.load()) pthread_join(g_reaper, nullptr);
+  if (env_flag("SOFA_LITE_DEBUG", false)) {
+    fprintf(stderr,
+            "[sofahsalite] queues=%lu batches=%lu kernel_pkts=%lu attached=%lu "
+            "skipped_has_signal=%lu reaped=%lu copy=%lu copy_engine=%lu "
+            "copy_rect=%lu copies_recorded=%lu pool_exhausted=%lu\n",
+            g_stats.queue_create_gpu.load(), g_stats.submit_batches.load(),
+            g_stats.kernel_pkts.load(), g_stats.attached.load(),
+            g_stats.skipped_has_signal.load(), g_stats.reaped.load(),
+            g_stats.copy_calls.load(), g_stats.copy_engine_calls.load(),
+            g_stats.copy_rect_calls.load(), g_stats.copies_recorded.load(),
+            g_pool_exhausted.load());
+  }
   if (g_out) {
     write_clock_rec();
     if (g_pool_exhausted.load() > 0) {
@@ -755,6 +809,7 @@ bool OnLoad(void* table_v, uint64_t runtime_version, uint64_t failed_tool_count,
     table->amd_ext_->hsa_amd_memory_async_copy_fn = async_copy_wrap;
     table->amd_ext_->hsa_amd_memory_async_copy_on_engine_fn =
         async_copy_engine_wrap;
+    table->amd_ext_->hsa_amd_memory_async_copy_rect_fn = async_copy_rect_wrap;
     // SDMA timestamps require async-copy profiling globally
     g_amd.hsa_amd_profiling_async_copy_enable_fn(true);
   }

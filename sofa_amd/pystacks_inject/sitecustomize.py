@@ -1,14 +1,16 @@
 """In-process Python stack sampler (pyflame replacement).
 
 The reference prefixes the target with `pyflame --flamechart -o pystacks.txt`
-(cyliustack/sofa bin/sofa_record.py:326-333); pyflame is dead/absent, so
-sofa_record injects this module via PYTHONPATH (sitecustomize imports
-automatically in every Python child) and a daemon thread samples
-sys._current_frames() at SOFA_PYSTACKS_HZ (default 50).
+(cyliustack/sofa bin/sofa_record.py:326-333); pyflame sampled the WHOLE
+interpreter, so this does too: every thread in sys._current_frames() is
+stacked per tick (round-1 verdict: main-thread-only sampling made dataloader
+and worker threads invisible).  Injected via PYTHONPATH (sitecustomize
+imports automatically in every Python child); a daemon thread samples at
+SOFA_PYSTACKS_HZ (default 50).
 
-Output format (one sample per two lines, matching the reference's
-pystacks.txt shape parsed at bin/sofa_preprocess.py:1709-1761):
-    <epoch seconds>
+Output format (one sample per two lines; header line grew tid + thread name
+fields, the parser accepts both shapes):
+    <epoch seconds> <tid> <thread-name>
     frameN;...;frame1;frame0
 """
 
@@ -30,22 +32,33 @@ if os.environ.get("SOFA_PYSTACKS_OUT"):
         except OSError:
             return
         period = 1.0 / max(_hz, 1.0)
-        main_thread = threading.main_thread().ident
+        own_ident = threading.get_ident()
+        names = {}
+        names_refresh = 0.0
         while True:
             t = time.time()
-            frames = sys._current_frames()
-            frame = frames.get(main_thread)
-            if frame is not None:
+            if t >= names_refresh:  # thread names change rarely; cache 1 s
+                names = {th.ident: th.name for th in threading.enumerate()}
+                names_refresh = t + 1.0
+            for ident, frame in sys._current_frames().items():
+                if ident == own_ident or frame is None:
+                    continue
                 stack = []
                 fr = frame
                 depth = 0
                 while fr is not None and depth < 64:
                     code = fr.f_code
-                    stack.append("%s (%s:%d)" % (code.co_name, os.path.basename(code.co_filename), fr.f_lineno))
+                    stack.append(
+                        "%s (%s:%d)"
+                        % (code.co_name, os.path.basename(code.co_filename), fr.f_lineno)
+                    )
                     fr = fr.f_back
                     depth += 1
-                if stack and "_sampler" not in stack[0]:
-                    f.write("%.6f\n%s\n" % (t, ";".join(reversed(stack))))
+                if stack:
+                    name = names.get(ident, "?").replace(" ", "_")
+                    f.write(
+                        "%.6f %d %s\n%s\n" % (t, ident, name, ";".join(reversed(stack)))
+                    )
             dt = period - (time.time() - t)
             if dt > 0:
                 time.sleep(dt)

@@ -180,6 +180,8 @@ def build_target_env(cfg: SofaConfig) -> dict:
                 prev = env.get("ROCP_TOOL_LIBRARIES", "")
                 env["ROCP_TOOL_LIBRARIES"] = tracer + ((":" + prev) if prev else "")
                 env["SOFA_TRACE_DISPATCH"] = "0"
+                env["SOFA_TRACE_COPY"] = "1"  # SDMA copies bypass the public
+                # HSA copy entries; the SDK's per-copy hook is ~free
                 env["SOFA_TRACE_RCCL"] = "1" if cfg.enable_rccl_trace else "0"
         elif os.path.exists(tracer):
             prev = env.get("ROCP_TOOL_LIBRARIES", "")

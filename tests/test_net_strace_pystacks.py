@@ -75,3 +75,52 @@ def test_parse_pystacks(tmp_path):
     assert (df["pid"] == 123).all()
     assert "<br>" in df["name"].iloc[0]
     assert np.isclose(df["timestamp"].iloc[0], 1.0)
+
+
+def test_parse_pystacks_multithread(tmp_path):
+    """tid-tagged headers: per-thread duration diffing + thread-name prefix."""
+    with open(os.path.join(tmp_path, "pystacks.txt.77"), "w") as f:
+        # main thread at t, t+0.02; worker at t+0.01, t+0.03
+        f.write("100.000000 111 MainThread\nmain (app.py:1);work (app.py:9)\n")
+        f.write("100.010000 222 loader_0\nrun (dl.py:5);fetch (dl.py:44)\n")
+        f.write("100.020000 111 MainThread\nmain (app.py:1);work (app.py:10)\n")
+        f.write("100.030000 222 loader_0\nrun (dl.py:5);decode (dl.py:60)\n")
+    df = parse_pystacks(str(tmp_path), None)
+    assert set(df["tid"].unique()) == {111, 222}
+    worker = df[df["tid"] == 222].sort_values("timestamp")
+    assert len(worker) == 2
+    # per-thread diff: 0.02 s between the two worker samples
+    assert abs(worker["duration"].iloc[0] - 0.02) < 1e-9
+    assert worker["name"].iloc[0].startswith("[loader_0] ")
+
+
+def test_live_sampler_sees_all_threads(tmp_path):
+    """End-to-end: inject the sampler into a child running two busy threads;
+    both must appear in the output (round-1 gap: only main was sampled)."""
+    import subprocess
+    import sys
+
+    inject = os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "sofa_amd",
+        "pystacks_inject",
+    )
+    code = (
+        "import threading, time\n"
+        "def spin(n):\n"
+        "    t0 = time.time()\n"
+        "    while time.time() - t0 < 0.8: n = (n * 7 + 1) % 1000003\n"
+        "th = threading.Thread(target=spin, args=(1,), name='worker_thread')\n"
+        "th.start(); spin(2); th.join()\n"
+    )
+    env = dict(os.environ)
+    env["PYTHONPATH"] = inject + os.pathsep + env.get("PYTHONPATH", "")
+    env["SOFA_PYSTACKS_OUT"] = os.path.join(str(tmp_path), "pystacks.txt")
+    env["SOFA_PYSTACKS_HZ"] = "100"
+    subprocess.run([sys.executable, "-c", code], env=env, timeout=60, check=True)
+    df = parse_pystacks(str(tmp_path), None)
+    assert len(df) > 10
+    names = df["name"].str.cat(sep="\n")
+    assert "worker_thread" in names, "dataloader-style worker thread not sampled"
+    assert "MainThread" in names
+    assert df["tid"].nunique() >= 2
