@@ -43,6 +43,9 @@ if os.environ.get("SOFA_PYSTACKS_OUT"):
             for ident, frame in sys._current_frames().items():
                 if ident == own_ident or frame is None:
                     continue
+                if ident not in names:  # new thread since last refresh
+                    names = {th.ident: th.name for th in threading.enumerate()}
+                    names_refresh = t + 1.0
                 stack = []
                 fr = frame
                 depth = 0
