@@ -42,21 +42,22 @@ def setup_tracer_env(logdir: str, tracer_mode: str) -> None:
     os.environ["SOFA_LOGDIR"] = logdir
     os.environ["SOFA_DEFER_START"] = "1"
     if tracer_mode == "lite":
-        # HSA-level dispatch/copy tracer + SDK collector demoted to
-        # RCCL-args only (sofa_amd/native/hsalite/hsalite.cc)
+        # HSA-level dispatch/copy tracer, SDK-free (rocprofiler's presence
+        # blocks hsalite's OnLoad — measured); RCCL args via debug log
         prev_hsa = os.environ.get("HSA_TOOLS_LIB", "")
         if LITE not in prev_hsa:
             os.environ["HSA_TOOLS_LIB"] = LITE + ((" " + prev_hsa) if prev_hsa else "")
-        os.environ["SOFA_TRACE_DISPATCH"] = "0"
-        os.environ["SOFA_TRACE_COPY"] = "1"
-        os.environ["SOFA_TRACE_RCCL"] = "1"
+        if "NCCL_DEBUG" not in os.environ:
+            os.environ["NCCL_DEBUG"] = "INFO"
+            os.environ["NCCL_DEBUG_SUBSYS"] = "COLL"
+            os.environ["NCCL_DEBUG_FILE"] = os.path.join(logdir, "rccl_debug.%h.%p")
     else:
         os.environ["SOFA_TRACE_HIP_API"] = "1"
         os.environ["SOFA_TRACE_RCCL"] = "1"
         os.environ.setdefault("SOFA_GPU_BUFFER_MB", "64")
-    prev = os.environ.get("ROCP_TOOL_LIBRARIES", "")
-    if TRACER not in prev:
-        os.environ["ROCP_TOOL_LIBRARIES"] = TRACER + ((":" + prev) if prev else "")
+        prev = os.environ.get("ROCP_TOOL_LIBRARIES", "")
+        if TRACER not in prev:
+            os.environ["ROCP_TOOL_LIBRARIES"] = TRACER + ((":" + prev) if prev else "")
 
 
 class TracerCtl:
@@ -64,13 +65,14 @@ class TracerCtl:
 
     def __init__(self, tracer_mode: str):
         self.libs = []
-        lib = ctypes.CDLL(TRACER)
-        lib.sofa_tracer_event_count.restype = ctypes.c_ulonglong
-        self.libs.append(("sofa_tracer", lib))
         if tracer_mode == "lite":
             ll = ctypes.CDLL(LITE)
             ll.sofa_lite_event_count.restype = ctypes.c_ulonglong
             self.libs.append(("sofa_lite", ll))
+        else:
+            lib = ctypes.CDLL(TRACER)
+            lib.sofa_tracer_event_count.restype = ctypes.c_ulonglong
+            self.libs.append(("sofa_tracer", lib))
 
     def start(self):
         for prefix, lib in self.libs:

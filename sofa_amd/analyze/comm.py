@@ -188,12 +188,16 @@ def attach_kernel_times(
         k_dur = kg["duration"].to_numpy(dtype=np.float64)
         order = np.argsort(k_ts, kind="stable")
         k_ts, k_dur = k_ts[order], k_dur[order]
+        # RCCL debug-log rows carry no wall-clock timestamps (lite mode,
+        # preprocess.rccl_log) — pure order matching, no time constraint
+        order_only = bool(np.all(api_ts[sel] < 1.0))
         ki = 0
         for idx in sel:
             # first unconsumed kernel starting at/after the API start
             # (10 us slack for clock-pair jitter between streams)
-            while ki < len(k_ts) and k_ts[ki] < api_ts[idx] - 10e-6:
-                ki += 1
+            if not order_only:
+                while ki < len(k_ts) and k_ts[ki] < api_ts[idx] - 10e-6:
+                    ki += 1
             if ki >= len(k_ts):
                 break
             kern_dur[idx] = k_dur[ki]

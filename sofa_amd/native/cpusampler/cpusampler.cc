@@ -333,6 +333,7 @@ int main(int argc, char** argv) {
   int freq = 99;
   pid_t target_pid = -1;  // -1 => system-wide
   bool system_wide = false;
+  const char* cgroup_path = nullptr;  // -G: scope to one cgroup (container)
   bool want_hw_cycles = false;  // -e cycles: try the PMU, fall back to sw
   uint64_t max_mb = 512;
   for (int i = 1; i < argc; i++) {
@@ -341,6 +342,7 @@ int main(int argc, char** argv) {
     else if (a == "-F" && i + 1 < argc) freq = atoi(argv[++i]);
     else if (a == "-p" && i + 1 < argc) target_pid = atoi(argv[++i]);
     else if (a == "-a") system_wide = true;
+    else if (a == "-G" && i + 1 < argc) cgroup_path = argv[++i];
     else if (a == "--max-mb" && i + 1 < argc) max_mb = strtoull(argv[++i], nullptr, 10);
     else if (a == "-g" || a == "--callchain") g_callchain = true;
     else if (a == "-e" && i + 1 < argc) {
@@ -351,9 +353,20 @@ int main(int argc, char** argv) {
       return 2;
     }
   }
-  if (!out_path || (!system_wide && target_pid < 0)) {
-    fprintf(stderr, "sofa-cpusampler: need -o and one of -a / -p\n");
+  if (!out_path || (!system_wide && target_pid < 0 && !cgroup_path)) {
+    fprintf(stderr, "sofa-cpusampler: need -o and one of -a / -p / -G\n");
     return 2;
+  }
+  // -G <cgroup dir>: profile only tasks in that cgroup (container target,
+  // reference `perf record --cgroup=docker/<cid>`, bin/sofa_record.py:394)
+  int cgroup_fd = -1;
+  if (cgroup_path) {
+    cgroup_fd = open(cgroup_path, O_RDONLY | O_DIRECTORY | O_CLOEXEC);
+    if (cgroup_fd < 0) {
+      fprintf(stderr, "sofa-cpusampler: open cgroup %s: %s\n", cgroup_path,
+              strerror(errno));
+      return 2;
+    }
   }
 
   signal(SIGTERM, on_signal);
@@ -431,8 +444,13 @@ int main(int argc, char** argv) {
   const size_t page = static_cast<size_t>(sysconf(_SC_PAGESIZE));
   for (int cpu = 0; cpu < n_cpus; cpu++) {
     Ring r;
-    r.fd = static_cast<int>(perf_event_open(
-        &attr, system_wide ? -1 : target_pid, cpu, -1, PERF_FLAG_FD_CLOEXEC));
+    unsigned long flags = PERF_FLAG_FD_CLOEXEC;
+    pid_t open_pid = system_wide ? -1 : target_pid;
+    if (cgroup_fd >= 0) {
+      open_pid = cgroup_fd;  // perf cgroup mode: pid = cgroup dir fd, per-cpu
+      flags |= PERF_FLAG_PID_CGROUP;
+    }
+    r.fd = static_cast<int>(perf_event_open(&attr, open_pid, cpu, -1, flags));
     if (r.fd < 0) {
       if (errno == ENODEV) continue;  // offline cpu
       fprintf(stderr, "sofa-cpusampler: perf_event_open cpu%d: %s\n", cpu,

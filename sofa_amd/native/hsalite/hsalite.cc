@@ -604,8 +604,10 @@ hsa_status_t async_copy_wrap(void* dst, hsa_agent_t dst_agent, const void* src,
                              const hsa_signal_t* dep_signals,
                              hsa_signal_t completion_signal) {
   g_stats.copy_calls.fetch_add(1, std::memory_order_relaxed);
-  if (g_mode != MODE_FULL || !g_armed.load(std::memory_order_relaxed) ||
-      completion_signal.handle == 0) {
+  // ROCclr passes completion_signal.handle == 0 on its SDMA path (measured
+  // on MI355X); a zero signal means nobody waits on it, so attaching ours is
+  // free of semantic risk — same rule as kernel packets.
+  if (g_mode != MODE_FULL || !g_armed.load(std::memory_order_relaxed)) {
     return g_amd.hsa_amd_memory_async_copy_fn(dst, dst_agent, src, src_agent,
                                               size, num_dep_signals,
                                               dep_signals, completion_signal);
@@ -646,8 +648,7 @@ hsa_status_t async_copy_engine_wrap(void* dst, hsa_agent_t dst_agent,
                                     hsa_amd_sdma_engine_id_t engine_id,
                                     bool force_copy_on_sdma) {
   g_stats.copy_engine_calls.fetch_add(1, std::memory_order_relaxed);
-  if (g_mode != MODE_FULL || !g_armed.load(std::memory_order_relaxed) ||
-      completion_signal.handle == 0) {
+  if (g_mode != MODE_FULL || !g_armed.load(std::memory_order_relaxed)) {
     return g_amd.hsa_amd_memory_async_copy_on_engine_fn(
         dst, dst_agent, src, src_agent, size, num_dep_signals, dep_signals,
         completion_signal, engine_id, force_copy_on_sdma);
@@ -778,9 +779,20 @@ int sofa_lite_active() { return g_out != nullptr; }
 
 unsigned long long sofa_lite_dropped() { return g_pool_exhausted.load(); }
 
+// SOFA_LITE_DEBUG=1 diagnostics: constructor fires iff ROCr dlopens us,
+// OnLoad print fires iff the tools ABI call happens — separates "never
+// loaded" from "loaded but not initialized" when other tools are present.
+__attribute__((constructor)) static void sofa_lite_ctor() {
+  if (env_flag("SOFA_LITE_DEBUG", false))
+    fprintf(stderr, "[sofahsalite] dlopened (pid %d)\n", getpid());
+}
+
 // ROCr tools-library entry points (HSA_TOOLS_LIB)
 bool OnLoad(void* table_v, uint64_t runtime_version, uint64_t failed_tool_count,
             const char* const* failed_tool_names) {
+  if (env_flag("SOFA_LITE_DEBUG", false))
+    fprintf(stderr, "[sofahsalite] OnLoad (runtime_version %lu)\n",
+            (unsigned long) runtime_version);
   (void) runtime_version;
   (void) failed_tool_count;
   (void) failed_tool_names;

@@ -178,6 +178,14 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
         write_trace_csv(df_gpu, os.path.join(logdir, "gputrace.csv"))
         traces.append(SOFATrace(name="gpu_traces", title="GPU kernels & copies", color="DarkSlateBlue", data=df_gpu))
         traces += _filter_traces(df_gpu, cfg.gpu_filters, "gpu")
+    if len(df_rccl) == 0:
+        # lite mode: collective args come from RCCL's own debug log
+        try:
+            from .rccl_log import parse_rccl_log
+
+            df_rccl = parse_rccl_log(logdir)
+        except Exception as e:
+            p.print_warning(f"rccl debug-log parse failed: {e}")
     if len(df_rccl):
         write_trace_csv(df_rccl, os.path.join(logdir, "rccltrace.csv"))
         traces.append(SOFATrace(name="rccl_traces", title="RCCL collectives", color="Crimson", data=df_rccl))

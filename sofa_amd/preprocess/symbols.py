@@ -189,7 +189,12 @@ class KernelSymbols:
 class Symbolizer:
     """Per-pid address space from sampler MMAP records + lazy DSO loading."""
 
-    def __init__(self, mmaps: Dict[int, List[Tuple[int, int, int, str]]], kallsyms: str = ""):
+    def __init__(
+        self,
+        mmaps: Dict[int, List[Tuple[int, int, int, str]]],
+        kallsyms: str = "",
+        container_root: str = "",
+    ):
         # pid -> sorted list of (start, end, pgoff, path)
         self.spaces: Dict[int, List[Tuple[int, int, int, str]]] = {}
         for pid, maps in mmaps.items():
@@ -197,6 +202,10 @@ class Symbolizer:
             self.spaces[pid] = entries
         self._dsos: Dict[str, DsoSymbols] = {}
         self.ksyms = KernelSymbols(kallsyms) if kallsyms else None
+        # in-container DSO paths resolve under the container overlayfs root
+        # recorded by record.docker_target (reference used a bindfs symfs,
+        # bin/sofa_preprocess.py:396-414)
+        self.container_root = container_root.rstrip("/")
 
     def _dso(self, path: str) -> DsoSymbols:
         if path not in self._dsos:
@@ -217,6 +226,10 @@ class Symbolizer:
             i = bisect.bisect_right(starts, ip) - 1
             if i >= 0:
                 start, end, pgoff, path = space[i]
+                if not os.path.isfile(path) and self.container_root:
+                    alt = self.container_root + path
+                    if os.path.isfile(alt):
+                        path = alt
                 if ip < end and os.path.isfile(path):
                     dso = self._dso(path)
                     file_addr = (ip - start + pgoff) if dso.pie else ip

@@ -256,3 +256,55 @@ def test_launch_latency_profile(tmp_path):
     assert 4.0 <= d["launch_latency_us_p50"] <= 6.0
     # 3us busy vs 91us gaps -> launch-bound ratio high
     assert d["gpu_idle_gap_ratio"] > 0.5
+
+
+RCCL_LOG_FIXTURE = """\
+node01:4242:4250 [0] NCCL INFO AllReduce: opCount 2a sendbuff 0x7f2a40000000 recvbuff 0x7f2a40000000 count 1048576 datatype 9 op 0 root 0 comm 0x55aabbcc [nranks=8] stream 0x55dd00 task 0 globalrank 0
+node01:4242:4250 [0] NCCL INFO AllGather: opCount 2b sendbuff 0x7f2a41000000 recvbuff 0x7f2a42000000 count 524288 datatype 7 op 0 root 0 comm 0x55aabbcc [nranks=8] stream 0x55dd00 task 0 globalrank 0
+node01:4242:4250 [0] NCCL INFO NET/Socket : some unrelated line
+"""
+
+
+def test_rccl_debug_log_parse(tmp_path):
+    """lite-mode collective args from RCCL's NCCL_DEBUG=INFO COLL channel."""
+    from sofa_amd.preprocess.rccl_log import parse_rccl_log
+
+    with open(os.path.join(tmp_path, "rccl_debug.node01.4242"), "w") as f:
+        f.write(RCCL_LOG_FIXTURE)
+    df = parse_rccl_log(str(tmp_path))
+    assert len(df) == 2
+    ar = df[df["name"].str.contains("ncclAllReduce")]
+    assert len(ar) == 1
+    assert ar["payload"].iloc[0] == 1048576 * 2  # bf16 (dtype 9)
+    ag = df[df["name"].str.contains("ncclAllGather")]
+    assert ag["payload"].iloc[0] == 524288 * 4  # f32 (dtype 7)
+    assert (df["copyKind"] == 16).all()
+    assert (df["deviceId"] == 0).all()
+    assert (df["pid"] == 4242).all()
+
+
+def test_rccl_log_order_matched_to_kernels(tmp_path):
+    """Log rows have no wall-clock timestamps; attach_kernel_times must
+    order-match them to ncclDevKernel spans."""
+    from sofa_amd.preprocess.rccl_log import parse_rccl_log
+
+    lines = []
+    for i in range(3):
+        lines.append(
+            "n:7:8 [2] NCCL INFO AllReduce: opCount %x sendbuff 0x1 recvbuff 0x1 "
+            "count 1000 datatype 7 op 0 root 0 comm 0xabc [nranks=4] stream 0x9 "
+            "task 0 globalrank 2" % i
+        )
+    with open(os.path.join(tmp_path, "rccl_debug.n.7"), "w") as f:
+        f.write("\n".join(lines) + "\n")
+    df_rccl = parse_rccl_log(str(tmp_path))
+    kern = new_trace_df(3)
+    kern["name"] = ["[gpu2] ncclDevKernel_AllReduce_Sum_f32_RING_LL"] * 3
+    kern["deviceId"] = 2
+    kern["copyKind"] = 0
+    kern["timestamp"] = [5.0, 5.1, 5.2]
+    kern["duration"] = [2e-3, 3e-3, 4e-3]
+    kern["pid"] = 7
+    dur, matched = comm_mod.attach_kernel_times(df_rccl, kern)
+    assert matched.all()
+    assert np.allclose(sorted(dur), [2e-3, 3e-3, 4e-3])
