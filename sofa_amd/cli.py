@@ -86,6 +86,10 @@ def build_parser() -> argparse.ArgumentParser:
         "the rocprofiler-sdk path",
     )
     ap.add_argument("--gpu_buffer_mb", type=int, default=64)
+    ap.add_argument("--docker", default="", dest="docker_image", metavar="IMAGE",
+                    help="profile inside a container of IMAGE (command = "
+                    "image CMD when omitted); cgroup-scoped CPU sampling + "
+                    "GPU tracer mounted read-only")
     ap.add_argument("--enable_blkio", action="store_true",
                     help="per-request block-IO latency via tracefs "
                     "block_rq_issue/complete (blktrace parity)")
@@ -161,6 +165,7 @@ def cfg_from_args(args) -> SofaConfig:
         gpu_tracer=args.gpu_tracer,
         enable_blkio=args.enable_blkio,
         blkdev=args.blkdev,
+        docker_image=args.docker_image,
         enable_rccl_trace=not args.no_rccl,
         rccl_shim=args.rccl_shim,
         gpu_ring_buffer_mb=args.gpu_buffer_mb,

@@ -97,6 +97,7 @@ class SofaConfig:
     gpu_ring_buffer_mb: int = 64       # collector buffer size per process
     blkdev: str = ""                   # block device for blktrace-like stats
     enable_blkio: bool = False         # tracefs block_rq_issue/complete per-IO tracing
+    docker_image: str = ""             # profile the command inside this container image
     nvsmi_interval_ms: int = 100       # GPU telemetry poll period
 
     # --- preprocess ---

@@ -71,7 +71,12 @@ def scs_to_cputrace(
         kallsyms = os.path.join(logdir, "kallsyms") if logdir else ""
         if kallsyms and not os.path.isfile(kallsyms):
             kallsyms = ""
-        symr = Symbolizer(scs.mmaps, kallsyms)
+        croot = ""
+        cr_path = os.path.join(logdir, "container_root.txt") if logdir else ""
+        if cr_path and os.path.isfile(cr_path):
+            with open(cr_path) as crf:
+                croot = crf.read().strip()
+        symr = Symbolizer(scs.mmaps, kallsyms, container_root=croot)
         # resolve UNIQUE (pid, ip, kernel-flag) triples only, then map back
         # vectorized (a long run has millions of samples but few unique IPs).
         # Structured key — no bit-packing, so large pids (pid_max can be

@@ -280,6 +280,20 @@ def sofa_record(command: str, cfg: SofaConfig, duration: float = 0.0) -> int:
     ensure_native_built(cfg.verbose)
     sofa_clean(cfg)
 
+    if getattr(cfg, "docker_image", ""):
+        # container target: image CMD (or `command`) profiled inside docker
+        # (reference bin/sofa_record.py:362-399)
+        with open(os.path.join(logdir, "sofa_time.txt"), "w") as f:
+            f.write("%.9f\n" % time.time())
+        _write_timebase(logdir)
+        try:
+            shutil.copyfile("/proc/kallsyms", os.path.join(logdir, "kallsyms"))
+        except OSError:
+            pass
+        from .docker_target import record_docker
+
+        return record_docker(cfg, cfg.docker_image, command, logdir)
+
     p.print_progress(f"recording into {logdir}: {command}")
 
     # --- prologue: clock base + symbols + topology ---
