@@ -162,6 +162,32 @@ def sofa_analyze(cfg: SofaConfig, pre: Optional[dict] = None) -> Dict[str, float
                     break
         except (OSError, ValueError, KeyError):
             pass
+        # MFMA-timed marker: the tracer's span for mfma_marker_kernel vs the
+        # kernel's own s_memrealtime measurement (on-device ground truth;
+        # north-star "MFMA-timed markers" validation)
+        try:
+            with open(tb_path) as f:
+                gtb = json.load(f)
+            self_ns = gtb.get("mfma_marker_self_ns", 0.0)
+            if self_ns > 0 and len(df_gpu):
+                marks = df_gpu[
+                    df_gpu["name"].astype(str).str.contains("mfma_marker_kernel")
+                ]
+                if len(marks):
+                    traced_ns = float(marks["duration"].max()) * 1e9
+                    err_pct = 100.0 * abs(traced_ns - self_ns) / self_ns
+                    features.append(("mfma_marker_clock_err_pct", float(err_pct)))
+                    msg = (
+                        "MFMA-timed marker: tracer span %.1f us vs on-device "
+                        "self-measurement %.1f us (%.2f%% error)"
+                        % (traced_ns / 1e3, self_ns / 1e3, err_pct)
+                    )
+                    if err_pct > 5.0:
+                        p.print_warning(msg)
+                    else:
+                        p.print_info(msg)
+        except (OSError, ValueError, KeyError):
+            pass
 
     # --- concurrency breakdown ---
     concurrency_breakdown(

@@ -60,6 +60,47 @@ uint64_t host_ns(clockid_t c) {
   return uint64_t(ts.tv_sec) * 1000000000ull + ts.tv_nsec;
 }
 
+// ---- MFMA-timed marker (BASELINE.json north star: "MFMA-timed markers") --
+//
+// A marker kernel whose duration is SELF-MEASURED on-device: one wave runs
+// `iters` chained v_mfma_f32_16x16x32_bf16 (gfx950 matrix-core op, 4
+// independent accumulator chains to keep the MFMA pipe fed) bracketed by
+// s_memrealtime stamps.  The tracer's reported span for this kernel can then
+// be cross-checked against the kernel's own tick count — a calibrated-
+// duration probe that validates collector timestamp scaling end-to-end
+// (the reference's cuhello.cu only *created* events; it could not check
+// the profiler's clock against ground truth).
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+__global__ void mfma_marker_kernel(uint64_t* out, int iters) {
+  int lane = threadIdx.x & 63;
+  bf16x8 a, b;
+  for (int i = 0; i < 8; ++i) {
+    a[i] = (__bf16)((lane + i) & 7);
+    b[i] = (__bf16)((lane - i) & 7);
+  }
+  f32x4 c0 = {0, 0, 0, 0}, c1 = {0, 0, 0, 0}, c2 = {0, 0, 0, 0},
+        c3 = {0, 0, 0, 0};
+  __syncthreads();
+  uint64_t t0 = __builtin_amdgcn_s_memrealtime();
+  for (int i = 0; i < iters; ++i) {
+    c0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c0, 0, 0, 0);
+    c1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c1, 0, 0, 0);
+    c2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c2, 0, 0, 0);
+    c3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c3, 0, 0, 0);
+  }
+  uint64_t t1 = __builtin_amdgcn_s_memrealtime();
+  if (lane == 0) {
+    out[0] = t0;
+    out[1] = t1;
+  }
+  // keep the accumulators live
+  float s = c0[0] + c1[1] + c2[2] + c3[3];
+  if (s == 1234567.0f) out[2] = 1;  // never taken; defeats DCE
+}
+
 }  // namespace
 
 extern "C" {
@@ -110,6 +151,36 @@ int sofa_gpu_timebase_freq(int device, int interval_ms, double* ticks_per_sec) {
   double dev_dt = double(dmin1 - dmin0);
   if (host_dt <= 0) return -1;
   *ticks_per_sec = dev_dt / (host_dt * 1e-9);
+  return 0;
+}
+
+// MFMA-timed marker: launches the self-timed matrix-core burst; fills the
+// device tick pair + the host MONOTONIC_RAW window around the launch, and
+// the implied on-device duration in ns (ticks * 10 at the CDNA 100 MHz
+// constant counter).  A tracer profiling this process will report a span for
+// `mfma_marker_kernel` that must agree with self_ns — the validation is done
+// by analyze (clock_validation feature).
+int sofa_gpu_mfma_marker(int device, int iters, uint64_t* host_before_ns,
+                         uint64_t* host_after_ns, uint64_t* self_ticks,
+                         double* self_ns) {
+  HIP_CHECK(hipSetDevice(device));
+  uint64_t* d_out = nullptr;
+  HIP_CHECK(hipMalloc(&d_out, 4 * sizeof(uint64_t)));
+  hipLaunchKernelGGL(mfma_marker_kernel, dim3(1), dim3(64), 0, 0, d_out, 16);
+  HIP_CHECK(hipDeviceSynchronize());  // warm: code-object load + MFMA clock-up
+
+  uint64_t before = host_ns(CLOCK_MONOTONIC_RAW);
+  hipLaunchKernelGGL(mfma_marker_kernel, dim3(1), dim3(64), 0, 0, d_out, iters);
+  HIP_CHECK(hipDeviceSynchronize());
+  uint64_t after = host_ns(CLOCK_MONOTONIC_RAW);
+
+  uint64_t h_out[2] = {0, 0};
+  HIP_CHECK(hipMemcpy(h_out, d_out, sizeof(h_out), hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(d_out));
+  *host_before_ns = before;
+  *host_after_ns = after;
+  *self_ticks = h_out[1] - h_out[0];
+  *self_ns = double(h_out[1] - h_out[0]) * 10.0;  // 100 MHz constant counter
   return 0;
 }
 

@@ -64,6 +64,33 @@ def sample_gpu_timebase(device: int = 0, rounds: int = 5) -> Optional[dict]:
     else:
         best["ticks_per_second"] = 1e8  # CDNA s_memrealtime nominal 100 MHz
     best["device"] = device
+
+    # MFMA-timed marker: a matrix-core burst that measures its own duration
+    # with s_memrealtime.  Any tracer profiling this process records a span
+    # for `mfma_marker_kernel`; analyze cross-checks that span against
+    # mfma_marker_self_ns (clock-scale validation against on-device ground
+    # truth; BASELINE.json north star "MFMA-timed markers").
+    try:
+        lib.sofa_gpu_mfma_marker.argtypes = [
+            ctypes.c_int, ctypes.c_int,
+            ctypes.POINTER(ctypes.c_uint64), ctypes.POINTER(ctypes.c_uint64),
+            ctypes.POINTER(ctypes.c_uint64), ctypes.POINTER(ctypes.c_double),
+        ]
+        hb = ctypes.c_uint64(0)
+        ha = ctypes.c_uint64(0)
+        ticks = ctypes.c_uint64(0)
+        self_ns = ctypes.c_double(0)
+        iters = 20000  # ~4 MFMA chains x 20k -> O(100 us) marker
+        if lib.sofa_gpu_mfma_marker(
+            device, iters, ctypes.byref(hb), ctypes.byref(ha),
+            ctypes.byref(ticks), ctypes.byref(self_ns),
+        ) == 0 and ticks.value > 0:
+            best["mfma_marker_iters"] = iters
+            best["mfma_marker_self_ticks"] = ticks.value
+            best["mfma_marker_self_ns"] = self_ns.value
+            best["mfma_marker_host_window_ns"] = ha.value - hb.value
+    except (OSError, AttributeError):
+        pass
     return best
 
 

@@ -307,10 +307,20 @@ def sofa_record(command: str, cfg: SofaConfig, duration: float = 0.0) -> int:
     if cfg.enable_gpu:
         dump_xgmi_topology(logdir)
         try:
-            from .gpu_timebase import write_gpu_timebase
-
-            if write_gpu_timebase(logdir):
-                p.print_info("GPU timebase microkernel correlation recorded")
+            # run the timebase/MFMA-marker prologue as a SUBPROCESS carrying
+            # the tracer env: the collector then records the
+            # mfma_marker_kernel span into the same logdir, and analyze can
+            # cross-check it against the marker's s_memrealtime
+            # self-measurement (on-device ground truth for the tracer clock)
+            rc = subprocess.run(
+                [sys.executable, "-c",
+                 "from sofa_amd.record.gpu_timebase import write_gpu_timebase;"
+                 f"import sys; sys.exit(0 if write_gpu_timebase({logdir!r}) else 1)"],
+                env=build_target_env(cfg), timeout=120,
+                cwd=os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__)))),
+            ).returncode
+            if rc == 0:
+                p.print_info("GPU timebase + MFMA marker correlation recorded")
         except Exception as e:
             p.print_warning(f"gpu timebase prologue failed: {e}")
 

@@ -70,3 +70,36 @@ def test_timebase_vs_rocprofiler_clock(lib):
     host_dt = h2 - h1
     dev_dt_ns = (d2 - d1) * 10  # 100 MHz ticks
     assert abs(dev_dt_ns - host_dt) / host_dt < 1e-2, (host_dt, dev_dt_ns)
+
+
+def _mfma(lib, iters):
+    import ctypes
+
+    lib.sofa_gpu_mfma_marker.argtypes = [
+        ctypes.c_int, ctypes.c_int,
+        ctypes.POINTER(ctypes.c_uint64), ctypes.POINTER(ctypes.c_uint64),
+        ctypes.POINTER(ctypes.c_uint64), ctypes.POINTER(ctypes.c_double),
+    ]
+    hb = ctypes.c_uint64(0)
+    ha = ctypes.c_uint64(0)
+    ticks = ctypes.c_uint64(0)
+    ns = ctypes.c_double(0)
+    rc = lib.sofa_gpu_mfma_marker(
+        0, iters, ctypes.byref(hb), ctypes.byref(ha),
+        ctypes.byref(ticks), ctypes.byref(ns),
+    )
+    assert rc == 0
+    return hb.value, ha.value, ticks.value, ns.value
+
+
+def test_mfma_marker_self_timing(lib):
+    """MFMA-timed marker (north star): the matrix-core burst's self-measured
+    duration must sit inside the host launch window and scale with iters."""
+    hb, ha, ticks, ns = _mfma(lib, 20000)
+    assert ticks > 0
+    assert ns < (ha - hb), "self-measured span exceeds host window"
+    assert ns > 1000, "marker too short to be a usable calibration span"
+    # linear scaling within 30% (clock ramps / scheduling noise tolerated)
+    _, _, ticks2, _ = _mfma(lib, 40000)
+    ratio = ticks2 / ticks
+    assert 1.4 < ratio < 2.6, f"iters scaling ratio {ratio}"
