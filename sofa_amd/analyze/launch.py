@@ -34,15 +34,33 @@ def launch_latency_profile(sgt_files, features: List[Tuple[str, float]]) -> None
         idx = np.searchsorted(a_sorted["corr_id"], k["corr_id"])
         idx = np.clip(idx, 0, len(a_sorted) - 1)
         matched = a_sorted["corr_id"][idx] == k["corr_id"]
-        if not matched.any():
-            continue
-        lat = (
-            k["start_ns"][matched].astype(np.int64)
-            - a_sorted["end_ns"][idx[matched]].astype(np.int64)
-        )
-        lat = lat[lat >= 0]  # negative = kernel started before API returned (async enqueue ahead)
-        if len(lat):
-            lat_all.append(lat)
+        if matched.any() and k["corr_id"].max() > 0:
+            lat = (
+                k["start_ns"][matched].astype(np.int64)
+                - a_sorted["end_ns"][idx[matched]].astype(np.int64)
+            )
+            lat = lat[lat >= 0]  # negative = kernel started before API returned
+            if len(lat):
+                lat_all.append(lat)
+        else:
+            # lite-collector traces: kernels carry no correlation ids; the
+            # host-side aqlSubmitBatch spans (op 60000, corr_id = packets in
+            # batch) are order-matched — expand each batch span over its
+            # packet count and zip with kernels in start order (exact for
+            # the common single-queue stream)
+            subs = a[a["op"] == 60000]
+            if len(subs):
+                order_s = np.argsort(subs["start_ns"])
+                ends = np.repeat(
+                    subs["end_ns"][order_s].astype(np.int64),
+                    np.maximum(subs["corr_id"][order_s].astype(np.int64), 1),
+                )
+                ks = np.sort(k, order="start_ns")
+                n = min(len(ends), len(ks))
+                lat = ks["start_ns"][:n].astype(np.int64) - ends[:n]
+                lat = lat[lat >= 0]
+                if len(lat):
+                    lat_all.append(lat)
         # device idle gaps between consecutive kernels on one queue
         ks = np.sort(k, order="start_ns")
         gaps = ks["start_ns"][1:].astype(np.int64) - ks["end_ns"][:-1].astype(np.int64)

@@ -129,6 +129,18 @@ void write_name_rec(uint16_t type, uint64_t id, const char* name, size_t len) {
   write_raw(buf.data(), total);
 }
 
+void write_opname_rec(uint32_t kind, uint32_t op, const char* name) {
+  size_t len = strlen(name);
+  size_t total = (sizeof(sgt::OpNameRec) + len + 1 + 7) & ~size_t(7);
+  std::vector<char> buf(total, 0);
+  auto* rec = reinterpret_cast<sgt::OpNameRec*>(buf.data());
+  rec->h = {sgt::REC_OPNAME, static_cast<uint16_t>(total), 0};
+  rec->kind = kind;
+  rec->op = op;
+  memcpy(buf.data() + sizeof(sgt::OpNameRec), name, len);
+  write_raw(buf.data(), total);
+}
+
 void write_clock_rec() {
   sgt::ClockRec rec{};
   rec.h = {sgt::REC_CLOCK, sizeof(sgt::ClockRec), 0};
@@ -231,6 +243,11 @@ std::vector<QueueCtx*> g_queues;  // never shrunk; ctx leak on destroy is fine
 
 constexpr uint32_t kSlotsPerQueue = 8192;
 
+// ApiRec op id for host-side AQL submit spans (preprocess resolves the name
+// through an OpNameRec written at init)
+constexpr uint32_t kAqlSubmitOp = 60000;
+bool g_submit_spans = true;  // SOFA_LITE_SUBMIT_SPANS=0 to disable
+
 std::atomic<uint64_t> g_pool_exhausted{0};
 
 // diagnostics (SOFA_LITE_DEBUG=1 prints at exit; sofa_lite_stats exports)
@@ -272,6 +289,7 @@ void on_submit(const void* pkts, uint64_t pkt_count, uint64_t /*user_pkt_index*/
     return;
   }
   g_stats.submit_batches.fetch_add(1, std::memory_order_relaxed);
+  uint64_t submit_t0 = g_submit_spans ? sys_now_ns() : 0;
   // find kernel dispatch packets we can instrument
   const auto* in = static_cast<const hsa_kernel_dispatch_packet_t*>(pkts);
   static thread_local std::vector<hsa_kernel_dispatch_packet_t> scratch;
@@ -322,6 +340,17 @@ void on_submit(const void* pkts, uint64_t pkt_count, uint64_t /*user_pkt_index*/
     g_stats.attached.fetch_add(1, std::memory_order_relaxed);
   }
   writer(patched ? (const void*) scratch.data() : pkts, pkt_count);
+  if (g_submit_spans) {
+    sgt::ApiRec rec{};
+    rec.h = {sgt::REC_HIPAPI, sizeof(sgt::ApiRec), 0};
+    rec.start_ns = submit_t0;
+    rec.end_ns = sys_now_ns();
+    rec.corr_id = pkt_count;  // packets in this batch
+    rec.tid = my_tid();
+    rec.op = kAqlSubmitOp;
+    write_raw(&rec, sizeof(rec));
+    g_n_records.fetch_add(1, std::memory_order_relaxed);
+  }
 }
 
 // Reaper: poll in-flight slots; on completion read HW timestamps, emit
@@ -478,6 +507,15 @@ hsa_status_t queue_create_wrap(hsa_agent_t agent, uint32_t size,
   return st;
 }
 
+void finalize_for_shutdown();
+
+hsa_status_t hsa_shut_down_wrap() {
+  // stop the reaper and close the file while signals are still alive; ROCr
+  // refcounts init/shutdown but one shutdown from the app is the exit path
+  finalize_for_shutdown();
+  return g_core.hsa_shut_down_fn();
+}
+
 hsa_status_t queue_destroy_wrap(hsa_queue_t* queue) {
   // drain: reaper keeps polling; just mark dead and let slots finish.
   // ROCr destroys the proxy after in-flight packets retire, so pending
@@ -549,6 +587,18 @@ struct CopySlot {
 
 std::mutex g_copy_mutex;
 std::vector<CopySlot*> g_copy_free;
+std::atomic<bool> g_copy_prof_enabled{false};
+
+// hsa_amd_profiling_async_copy_enable cannot run from OnLoad (mid-hsa_init);
+// enable lazily at the first wrapped copy call
+void ensure_copy_profiling() {
+  bool expected = false;
+  if (g_copy_prof_enabled.compare_exchange_strong(expected, true)) {
+    hsa_status_t st = g_amd.hsa_amd_profiling_async_copy_enable_fn(true);
+    if (st != HSA_STATUS_SUCCESS && env_flag("SOFA_LITE_DEBUG", false))
+      fprintf(stderr, "[sofahsalite] async_copy_enable failed: %d\n", st);
+  }
+}
 
 CopySlot* copy_slot_get() {
   std::lock_guard<std::mutex> lk(g_copy_mutex);
@@ -571,10 +621,16 @@ void copy_slot_put(CopySlot* s) {
   g_copy_free.push_back(s);
 }
 
+std::atomic<uint64_t> g_copy_handler_fired{0};
+std::atomic<uint64_t> g_copy_time_fail{0};
+
 bool copy_done_handler(hsa_signal_value_t, void* arg) {
+  g_copy_handler_fired.fetch_add(1, std::memory_order_relaxed);
   auto* s = static_cast<CopySlot*>(arg);
   hsa_amd_profiling_async_copy_time_t t{};
   hsa_status_t st = g_amd.hsa_amd_profiling_get_async_copy_time_fn(s->sig, &t);
+  if (st != HSA_STATUS_SUCCESS)
+    g_copy_time_fail.fetch_add(1, std::memory_order_relaxed);
   // forward the caller's completion FIRST (its waiters matter more than our
   // record)
   if (s->orig_sig.handle != 0)
@@ -613,6 +669,7 @@ hsa_status_t async_copy_wrap(void* dst, hsa_agent_t dst_agent, const void* src,
                                               dep_signals, completion_signal);
   }
   enumerate_agents();
+  ensure_copy_profiling();
   CopySlot* s = copy_slot_get();
   if (!s)
     return g_amd.hsa_amd_memory_async_copy_fn(dst, dst_agent, src, src_agent,
@@ -654,6 +711,7 @@ hsa_status_t async_copy_engine_wrap(void* dst, hsa_agent_t dst_agent,
         completion_signal, engine_id, force_copy_on_sdma);
   }
   enumerate_agents();
+  ensure_copy_profiling();
   CopySlot* s = copy_slot_get();
   if (!s)
     return g_amd.hsa_amd_memory_async_copy_on_engine_fn(
@@ -719,8 +777,12 @@ void open_output() {
   hdr.monotonic_raw_ns = host_ns(CLOCK_MONOTONIC_RAW);
   hdr.rocp_ns = sys_now_ns();
   fwrite(&hdr, sizeof(hdr), 1, g_out);
+  write_opname_rec(3 /* HIP_RUNTIME_API kind */, 60000, "aqlSubmitBatch");
   write_clock_rec();
 }
+
+void finalize();
+void finalize_for_shutdown() { finalize(); }
 
 void finalize() {
   static std::atomic<bool> done{false};
@@ -739,6 +801,8 @@ void finalize() {
             g_stats.copy_calls.load(), g_stats.copy_engine_calls.load(),
             g_stats.copy_rect_calls.load(), g_stats.copies_recorded.load(),
             g_pool_exhausted.load());
+    fprintf(stderr, "[sofahsalite] copy_handler_fired=%lu copy_time_fail=%lu\n",
+            g_copy_handler_fired.load(), g_copy_time_fail.load());
   }
   if (g_out) {
     write_clock_rec();
@@ -809,6 +873,7 @@ bool OnLoad(void* table_v, uint64_t runtime_version, uint64_t failed_tool_count,
     else if (!strcmp(mode, "prof")) g_mode = MODE_PROF;
   }
   g_replace_signals = env_flag("SOFA_LITE_REPLACE_SIGNALS", false);
+  g_submit_spans = env_flag("SOFA_LITE_SUBMIT_SPANS", true);
   g_armed.store(!env_flag("SOFA_DEFER_START", false));
 
   open_output();
@@ -817,13 +882,12 @@ bool OnLoad(void* table_v, uint64_t runtime_version, uint64_t failed_tool_count,
   table->core_->hsa_queue_create_fn = queue_create_wrap;
   table->core_->hsa_queue_destroy_fn = queue_destroy_wrap;
   table->core_->hsa_executable_freeze_fn = exe_freeze_wrap;
+  table->core_->hsa_shut_down_fn = hsa_shut_down_wrap;
   if (g_mode == MODE_FULL) {
     table->amd_ext_->hsa_amd_memory_async_copy_fn = async_copy_wrap;
     table->amd_ext_->hsa_amd_memory_async_copy_on_engine_fn =
         async_copy_engine_wrap;
     table->amd_ext_->hsa_amd_memory_async_copy_rect_fn = async_copy_rect_wrap;
-    // SDMA timestamps require async-copy profiling globally
-    g_amd.hsa_amd_profiling_async_copy_enable_fn(true);
   }
   atexit(finalize);
   return true;

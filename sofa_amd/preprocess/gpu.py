@@ -268,7 +268,8 @@ def sgt_to_hip_api_trace(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.Dat
         hip_names = {
             op: nm
             for (kind, op), nm in sgt.opnames.items()
-            if nm.startswith("hip")
+            # "aql*": the lite collector's host-side submit spans
+            if nm.startswith("hip") or nm.startswith("aql")
         }
         op = a["op"]
         uniq, inv = np.unique(op, return_inverse=True)
