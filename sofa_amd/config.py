@@ -84,11 +84,13 @@ class SofaConfig:
     # copies/syncs/allocs — measured ~0 extra overhead at 2x events/s);
     # full API tracing (--hip_api_full) costs ~45% on launch-dense steps
     # (profiles/overhead_matrix_r01.md)
-    # "sdk" = rocprofiler-sdk collector (full fidelity: HIP API spans, KFD,
-    # allocs); "lite" = HSA-level dispatch tracer (lowest overhead) + SDK
-    # demoted to RCCL-args/markers only.  Flipped to lite-by-default once
-    # GPU-validated.
-    gpu_tracer: str = "sdk"
+    # "lite" (default) = HSA-level dispatch/copy tracer, 3.4% measured
+    # overhead on ResNet-50 bs=64 (profiles/overhead_decomp_r02.md), RCCL
+    # args via debug-log channel; "sdk" = rocprofiler-sdk collector (full
+    # fidelity: HIP API spans, RCCL args with corr ids, KFD events, allocs)
+    # at ~13% overhead.  The two cannot share a process (ROCr skips
+    # HSA_TOOLS_LIB tools when rocprofiler is registered - measured).
+    gpu_tracer: str = "lite"
     enable_gpu_hip_api: bool = True
     hip_api_full: bool = False
     enable_rccl_trace: bool = True     # RCCL API tracing via collector

@@ -117,6 +117,7 @@ def build_all(verbose: bool = False) -> None:
 
     # --- HIP device code (gfx950) ---
     hipcc = os.path.join(ROCM, "bin", "hipcc")
+    ring_h = os.path.join(HERE, "hip", "trace_ring.h")
     hip_targets = [
         (
             os.path.join(LIB, "libsofahip.so"),
@@ -124,11 +125,12 @@ def build_all(verbose: bool = False) -> None:
                 os.path.join(HERE, "hip", "timebase_kernel.hip"),
                 os.path.join(HERE, "hip", "trace_ring.hip"),
                 os.path.join(HERE, "hip", "ring_writer.hip"),
+                ring_h,
             ],
         ),
         (
             os.path.join(BIN, "sofa-bandwidth"),
-            [os.path.join(HERE, "hip", "bandwidth.hip")],
+            [os.path.join(HERE, "hip", "bandwidth.hip"), ring_h],
         ),
     ]
     for dst, srcs in hip_targets:
@@ -145,7 +147,10 @@ def build_all(verbose: bool = False) -> None:
         ]
         if dst.endswith(".so"):
             cmd += ["-fPIC", "-shared"]
-        cmd += ["-o", dst] + srcs
+        cmd += ["-o", dst] + [s for s in srcs if s.endswith(".hip")]
+        if dst.endswith("sofa-bandwidth"):
+            # --ring mode links the device-trace-ring machinery
+            cmd += ["-L" + LIB, "-lsofahip", "-Wl,-rpath,$ORIGIN/../lib"]
         _run(cmd, verbose)
 
 

@@ -34,37 +34,9 @@
     }                                                                         \
   } while (0)
 
+#include "trace_ring.h"
+
 namespace {
-
-struct __align__(16) RingRec {
-  uint64_t t_start;  // device ticks (s_memrealtime) or pre-converted ns
-  uint64_t t_end;
-  uint32_t tag;      // event class; 0 = empty slot
-  uint32_t src;      // producer id (wave/block/stream tag)
-  uint64_t arg;      // user payload
-};
-static_assert(sizeof(RingRec) == 32, "RingRec must be 32 bytes");
-
-struct RingControl {
-  unsigned long long head;  // total pushes (monotonic)
-  uint32_t capacity;
-  uint32_t _pad;
-};
-
-// ---------------------------------------------------------------- device API
-
-__device__ inline void ring_push(RingControl* ctl, RingRec* slots, uint32_t tag,
-                                 uint32_t src, uint64_t arg, uint64_t t0,
-                                 uint64_t t1) {
-  unsigned long long h = atomicAdd(&ctl->head, 1ull);
-  RingRec r;
-  r.t_start = t0;
-  r.t_end = t1;
-  r.tag = tag;
-  r.src = src;
-  r.arg = arg;
-  slots[h % ctl->capacity] = r;
-}
 
 // Test producer: each thread pushes one deterministic record.  Mirrored by
 // the CPU reference in tests/test_gpu_trace_ring.py.
@@ -187,6 +159,16 @@ int sofa_ring_create(int device, uint32_t capacity, void** ring_out) {
   HIP_CHECK(hipMemcpy(ring->d_ctl, &ctl, sizeof(ctl), hipMemcpyHostToDevice));
   HIP_CHECK(hipMemset(ring->d_slots, 0, sizeof(RingRec) * (size_t) capacity));
   *ring_out = ring;
+  return 0;
+}
+
+// Device-side pointers for EXTERNAL producers (instrumented kernels link
+// against libsofahip and pass these to their own launches — the product
+// path; see bandwidth.hip --ring)
+int sofa_ring_device_ptrs(void* ring_p, void** ctl_out, void** slots_out) {
+  Ring* ring = static_cast<Ring*>(ring_p);
+  *ctl_out = ring->d_ctl;
+  *slots_out = ring->d_slots;
   return 0;
 }
 

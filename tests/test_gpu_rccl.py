@@ -33,7 +33,8 @@ def test_rccl_allreduce_traced(tmp_path):
         "print('ar-done')\n"
     )
     r = subprocess.run(
-        [sys.executable, SOFA, "stat", f"{sys.executable} {snippet}", "--logdir", logdir],
+        [sys.executable, SOFA, "stat", f"{sys.executable} {snippet}", "--logdir", logdir,
+         "--gpu_tracer", "sdk"],
         capture_output=True, text=True, timeout=900,
     )
     assert "Complete!!" in r.stdout, (r.stdout[-2000:], r.stderr[-2000:])
