@@ -176,6 +176,29 @@ def diskstat_profile(logdir: str, features: Features) -> None:
     features.append(("disk_write_Bps", float(d["write_Bps"].mean())))
 
 
+def xgmi_measured_profile(logdir: str, features: Features) -> None:
+    """Measured per-xGMI-link bandwidth from gpu_metrics HW accumulators
+    (xgmi_counters.csv) — the ground truth the analytic ring model in
+    comm.rccl_link_attribution estimates; report both so a model/HW mismatch
+    is visible (each MI355X link peaks ~153 GB/s)."""
+    path = os.path.join(logdir, "xgmi_counters.csv")
+    if not os.path.isfile(path):
+        return
+    try:
+        d = pd.read_csv(path)
+    except (OSError, pd.errors.ParserError):
+        return
+    if len(d) == 0:
+        return
+    print("\nMeasured xGMI link bandwidth (gpu_metrics accumulators):")
+    print("%-6s %-5s %-6s %10s %10s" % ("gpu", "link", "dir", "mean GB/s", "max GB/s"))
+    for (dev, link, kind), grp in d.groupby(["dev", "link", "kind"]):
+        print("%-6d %-5d %-6s %10.2f %10.2f"
+              % (dev, link, kind, grp["GBps"].mean(), grp["GBps"].max()))
+    features.append(("xgmi_meas_max_GBps", float(d["GBps"].max())))
+    features.append(("xgmi_meas_links_active", float(d.groupby(["dev", "link"]).ngroups)))
+
+
 def blkio_latency_profile(df_blk: pd.DataFrame, features: Features) -> None:
     """Per-IO latency summary per device (reference blktrace_latency_profile,
     bin/sofa_analyze.py:596-638 — btt replaced by our matched issue/complete

@@ -158,9 +158,22 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
     t_sm, t_gmem, gpusmi_csv = sysmon.parse_gpusmi(logdir, tb)
     if len(t_sm):
         traces.append(SOFATrace(name="gpusmi_sm_traces", title="GPU busy (%)", color="DarkOrange", data=t_sm))
-        traces.append(SOFATrace(name="gpusmi_mem_traces", title="GPU mem busy (%)", color="Gold", data=t_gmem))
+        traces.append(SOFATrace(name="gpusmi_mem_traces", title="GPU mem/MM busy (%)", color="Gold", data=t_gmem))
         gpusmi_csv.to_csv(os.path.join(logdir, "gpusmi_trace.csv"), index=False)
     result["df_gpusmi"] = t_sm
+
+    # measured per-xGMI-link bandwidth (gpu_metrics HW accumulators)
+    try:
+        t_xgmi, xgmi_csv = sysmon.parse_xgmi_counters(logdir, tb)
+        if len(t_xgmi):
+            traces.append(
+                SOFATrace(name="xgmi_link_traces", title="xGMI link BW (GB/s)", color="MediumOrchid", data=t_xgmi)
+            )
+            xgmi_csv.to_csv(os.path.join(logdir, "xgmi_counters.csv"), index=False)
+        result["df_xgmi"] = t_xgmi
+    except Exception as e:
+        p.print_warning(f"xgmi counter parse failed: {e}")
+        result["df_xgmi"] = new_trace_df(0)
 
     # ---------------- GPU activity ----------------
     df_gpu = new_trace_df(0)

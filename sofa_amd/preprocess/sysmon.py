@@ -211,12 +211,25 @@ def parse_gpusmi(logdir: str, tb: Optional[TimeBase]):
     """gpusmi.txt -> (sm_trace, mem_trace, gpusmi_csv).
 
     Mirrors the reference's nvsmi trace semantics (bin/sofa_preprocess.py:
-    1013-1089): event 0 = GPU busy %, event 1 = memory busy %.
+    1013-1089): event 0 = GPU busy %, event 1 = memory busy %, event 2 = MM
+    (media/VCN) engine busy % — the dmon enc/dec analog (:1097-1183); rows
+    with -1 MM (no gpu_metrics support) drop the MM series.  Older 6-column
+    files (no mm) still parse.
     """
+    path = os.path.join(logdir, "gpusmi.txt")
+    ncols = 0
+    if os.path.isfile(path):
+        with open(path) as f:
+            first = f.readline().split()
+            ncols = len(first)
     cols = ["ts", "dev", "busy", "membusy", "vram", "power"]
+    if ncols >= 7:
+        cols = cols + ["mm"]
     raw = _read(logdir, "gpusmi.txt", cols)
     if raw is None:
         return new_trace_df(0), new_trace_df(0), None
+    if "mm" not in raw.columns:
+        raw["mm"] = -1
     d = raw[raw["busy"] >= 0]
     if len(d) == 0:
         return new_trace_df(0), new_trace_df(0), None
@@ -239,8 +252,73 @@ def parse_gpusmi(logdir: str, tb: Optional[TimeBase]):
         "gpu%d_mem:%d%%" % (dev, b) for dev, b in zip(d["dev"], d["membusy"])
     ]
 
+    mm_rows = d[d["mm"] >= 0]
+    if len(mm_rows):
+        mmt = new_trace_df(len(mm_rows))
+        mmt["timestamp"] = _tl(tb, mm_rows["ts"])
+        mmt["event"] = 2.0
+        mmt["duration"] = mm_rows["mm"].to_numpy(dtype=np.float64)
+        mmt["deviceId"] = mm_rows["dev"].to_numpy(dtype=np.int64)
+        mmt["name"] = [
+            "gpu%d_mm:%d%%" % (dev, b) for dev, b in zip(mm_rows["dev"], mm_rows["mm"])
+        ]
+        mem = pd.concat([mem, mmt], ignore_index=True)
+
     csv = d.copy()
     csv["timestamp"] = _tl(tb, d["ts"])
     csv["vram_MB"] = d["vram"].clip(lower=0) / 1e6
     csv["power_W"] = d["power"].clip(lower=0)
     return sm, mem, csv
+
+
+def parse_xgmi_counters(logdir: str, tb: Optional[TimeBase]):
+    """xgmi_counters.txt (per-device HW accumulators: ts dev r0..r7 w0..w7,
+    KB) -> per-link measured bandwidth rows + xgmi_counters.csv dataframe.
+
+    This is the MEASURED counterpart of analyze.comm's analytic ring model:
+    deltas of the gpu_metrics xgmi_read/write_data_acc counters over the poll
+    interval give true per-link GB/s regardless of which collective/algorithm
+    produced the traffic.
+    """
+    cols = ["ts", "dev"] + ["r%d" % i for i in range(8)] + ["w%d" % i for i in range(8)]
+    raw = _read(logdir, "xgmi_counters.txt", cols)
+    if raw is None or len(raw) < 2:
+        return new_trace_df(0), None
+    frames = []
+    csv_rows = []
+    for dev, grp in raw.groupby("dev"):
+        grp = grp.sort_values("ts")
+        dt = grp["ts"].diff().to_numpy()
+        for link in range(8):
+            for kind, pref in (("read", "r"), ("write", "w")):
+                acc = grp[f"{pref}{link}"].to_numpy(dtype=np.float64)
+                delta_kb = np.diff(acc, prepend=acc[0])
+                with np.errstate(divide="ignore", invalid="ignore"):
+                    gbps = np.where(
+                        (dt > 0) & (delta_kb > 0), delta_kb * 1e3 / np.maximum(dt, 1e-9) / 1e9, 0.0
+                    )
+                sel = gbps > 0.01  # suppress idle-link noise rows
+                if not sel.any():
+                    continue
+                n = int(sel.sum())
+                t = new_trace_df(n)
+                t["timestamp"] = _tl(tb, grp["ts"][sel])
+                t["duration"] = np.maximum(dt[sel], 0)
+                t["deviceId"] = int(dev)
+                t["event"] = float(link)
+                t["bandwidth"] = gbps[sel] * 1e9
+                t["payload"] = (delta_kb[sel] * 1e3).astype(np.int64)
+                t["pkt_src"] = int(dev)
+                t["pkt_dst"] = link
+                t["name"] = [
+                    "xgmi gpu%d link%d %s %.2f GB/s" % (dev, link, kind, g)
+                    for g in gbps[sel]
+                ]
+                frames.append(t)
+                for ts_v, g in zip(grp["ts"][sel], gbps[sel]):
+                    csv_rows.append((ts_v, int(dev), link, kind, g))
+    if not frames:
+        return new_trace_df(0), None
+    trace = pd.concat(frames, ignore_index=True).sort_values("timestamp")
+    csv = pd.DataFrame(csv_rows, columns=["ts", "dev", "link", "kind", "GBps"])
+    return trace.reset_index(drop=True), csv

@@ -132,6 +132,7 @@ class SysMonitor(threading.Thread):
         f_net = self._open("netstat.txt")
         f_vm = self._open("vmstat.txt")
         f_gpu = self._open("gpusmi.txt")
+        f_xgmi = self._open("xgmi_counters.txt")
         if self.enable_gpu:
             try:
                 self.smi = RocmSmi()
@@ -172,8 +173,10 @@ class SysMonitor(threading.Thread):
                     membusy = self.smi.memory_busy_percent(dev)
                     vram = self.smi.memory_usage(dev)
                     power = self.smi.power_watts(dev)
+                    gm = self.smi.gpu_metrics(dev)
+                    mm = gm["mm_activity"] if gm else -1
                     f_gpu.write(
-                        "%.6f %d %d %d %d %.1f\n"
+                        "%.6f %d %d %d %d %.1f %d\n"
                         % (
                             ts,
                             dev,
@@ -181,8 +184,21 @@ class SysMonitor(threading.Thread):
                             membusy if membusy is not None else -1,
                             vram if vram is not None else -1,
                             power if power is not None else -1.0,
+                            mm,
                         )
                     )
+                    # per-xGMI-link HW traffic accumulators (KB): measured
+                    # ground truth the analytic ring model reconciles against
+                    if gm and (any(gm["xgmi_read_kb"]) or any(gm["xgmi_write_kb"])):
+                        f_xgmi.write(
+                            "%.6f %d %s %s\n"
+                            % (
+                                ts,
+                                dev,
+                                " ".join(str(v) for v in gm["xgmi_read_kb"]),
+                                " ".join(str(v) for v in gm["xgmi_write_kb"]),
+                            )
+                        )
 
         next_t = time.time()
         while not self.stop_event.is_set():

@@ -131,6 +131,47 @@ class RocmSmi:
             pass
         return None
 
+    # rsmi_gpu_metrics_t field offsets, verified against the installed
+    # header with a C offsetof probe (rocm_smi.h:1051-1244, ROCm 7.2):
+    # the driver fills structure_size at +0; fields beyond it stay zero.
+    _GM_SIZE = 4544
+    _GM_GFX = 16       # u16 average_gfx_activity (%)
+    _GM_UMC = 18       # u16 average_umc_activity (%)
+    _GM_MM = 20        # u16 average_mm_activity (%) — VCN enc/dec, the
+    #                    nvsmi-dmon enc/dec analog (ref preprocess:1097-1183)
+    _GM_VCN = 122      # u16[4] vcn_activity
+    _GM_XGMI_R = 184   # u64[8] xgmi_read_data_acc (KB, accumulated)
+    _GM_XGMI_W = 248   # u64[8] xgmi_write_data_acc
+    _GM_JPEG = 352     # u16[32] jpeg_activity
+
+    def gpu_metrics(self, dev: int) -> Optional[dict]:
+        """Raw gpu_metrics table: media-engine busy + per-xGMI-link traffic
+        accumulators (the HW counters behind the analytic ring model)."""
+        if not self.available:
+            return None
+        import struct as _struct
+
+        buf = (ctypes.c_uint8 * self._GM_SIZE)()
+        fn = getattr(self.lib, "rsmi_dev_gpu_metrics_info_get", None)
+        if fn is None or fn(dev, ctypes.byref(buf)) != RSMI_STATUS_SUCCESS:
+            return None
+        raw = bytes(buf)
+        gfx, umc, mm = _struct.unpack_from("<HHH", raw, self._GM_GFX)
+        vcn = _struct.unpack_from("<4H", raw, self._GM_VCN)
+        xr = _struct.unpack_from("<8Q", raw, self._GM_XGMI_R)
+        xw = _struct.unpack_from("<8Q", raw, self._GM_XGMI_W)
+        jpeg = _struct.unpack_from("<32H", raw, self._GM_JPEG)
+        invalid = 0xFFFF
+        return {
+            "gfx_activity": gfx if gfx != invalid else -1,
+            "umc_activity": umc if umc != invalid else -1,
+            "mm_activity": mm if mm != invalid else -1,
+            "vcn_activity": [v if v != invalid else -1 for v in vcn],
+            "jpeg_activity": [v if v != invalid else -1 for v in jpeg],
+            "xgmi_read_kb": list(xr),
+            "xgmi_write_kb": list(xw),
+        }
+
     def topology_matrix(self) -> List[List[dict]]:
         """NxN matrix of {hops,type,weight,bw_min,bw_max} between GPU pairs."""
         n = self.n_devices

@@ -95,3 +95,61 @@ def test_missing_files_return_empty(tmp_path):
     assert len(tdf) == 0
     sm, mem, csv = sysmon.parse_gpusmi(str(tmp_path), TB)
     assert len(sm) == 0
+
+
+def test_parse_gpusmi_mm_column(tmp_path):
+    """7-column gpusmi (with MM/media engine busy) -> event-2 series; the
+    nvsmi dmon enc/dec analog (reference bin/sofa_preprocess.py:1097-1183)."""
+    with open(os.path.join(tmp_path, "gpusmi.txt"), "w") as f:
+        f.write("100.0 0 80 40 1000000 500.0 25\n")
+        f.write("100.1 0 85 45 1000000 505.0 30\n")
+    sm, mem, csv = sysmon.parse_gpusmi(str(tmp_path), None)
+    assert len(sm) == 2
+    mm = mem[mem["event"] == 2.0]
+    assert len(mm) == 2
+    assert list(mm["duration"]) == [25.0, 30.0]
+    assert "gpu0_mm:25%" in mm["name"].iloc[0]
+
+
+def test_parse_gpusmi_legacy_6col(tmp_path):
+    with open(os.path.join(tmp_path, "gpusmi.txt"), "w") as f:
+        f.write("100.0 0 80 40 1000000 500.0\n")
+    sm, mem, csv = sysmon.parse_gpusmi(str(tmp_path), None)
+    assert len(sm) == 1
+    assert (mem["event"] == 2.0).sum() == 0  # no MM series
+
+
+def test_parse_xgmi_counters(tmp_path):
+    """HW accumulator deltas -> per-link GB/s (measured ground truth for the
+    analytic ring model)."""
+    r0 = [0] * 8
+    r1 = [0] * 8
+    r1[3] = 15_000_000  # link 3: 15 GB read over 0.1 s -> 150 GB/s
+    w1 = [0] * 8
+    w1[3] = 1_000_000
+    with open(os.path.join(tmp_path, "xgmi_counters.txt"), "w") as f:
+        f.write("100.0 0 %s %s\n" % (" ".join(map(str, r0)), " ".join(map(str, r0))))
+        f.write("100.1 0 %s %s\n" % (" ".join(map(str, r1)), " ".join(map(str, w1))))
+    trace, csv = sysmon.parse_xgmi_counters(str(tmp_path), None)
+    assert len(trace) == 2  # read + write rows on link 3
+    rd = csv[(csv["kind"] == "read") & (csv["link"] == 3)]
+    assert len(rd) == 1
+    assert abs(rd["GBps"].iloc[0] - 150.0) < 1.0
+    wr = csv[csv["kind"] == "write"]
+    assert abs(wr["GBps"].iloc[0] - 10.0) < 0.2
+
+
+def test_xgmi_measured_profile_features(tmp_path):
+    import pandas as pd
+
+    from sofa_amd.analyze import profiles
+
+    pd.DataFrame(
+        {"ts": [1.0, 1.1], "dev": [0, 0], "link": [3, 3],
+         "kind": ["read", "read"], "GBps": [120.0, 140.0]}
+    ).to_csv(os.path.join(tmp_path, "xgmi_counters.csv"), index=False)
+    feats = []
+    profiles.xgmi_measured_profile(str(tmp_path), feats)
+    d = dict(feats)
+    assert d["xgmi_meas_max_GBps"] == 140.0
+    assert d["xgmi_meas_links_active"] == 1.0
