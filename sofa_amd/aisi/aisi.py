@@ -152,6 +152,22 @@ def sofa_aisi(logdir, cfg, df_cpu, df_gpu, df_rccl, features,
         "detected %d iterations (pattern length %d kernels); mean step %.4f s"
         % (len(idf), plen, idf["step_time"].mean())
     )
+    # expose the mined kernel-name sequence itself (round-1 roadmap: the
+    # pattern was detected but never shown); full sequence -> artifact,
+    # compressed head -> console
+    try:
+        pat_tokens = tokens[occ[0] : occ[0] + plen]
+        pat_names = [vocab.get(t, str(t)) for t in pat_tokens]
+        with open(os.path.join(logdir, "iteration_pattern.txt"), "w") as f:
+            for i, nm in enumerate(pat_names):
+                f.write("%4d %s\n" % (i, nm))
+        shown = [nm if len(nm) < 60 else nm[:57] + "..." for nm in pat_names[:12]]
+        print("iteration pattern (first %d of %d kernels; full list in "
+              "iteration_pattern.txt):" % (len(shown), plen))
+        for nm in shown:
+            print("    " + nm)
+    except (OSError, IndexError):
+        pass
     print(
         idf[["t_begin", "step_time", "fw_time", "bw_time", "gemm_time", "copy_time", "coll_time"]]
         .describe()

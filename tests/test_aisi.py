@@ -181,3 +181,23 @@ def test_suffix_automaton_counts_property():
                 )
                 assert brute == cnt, (tokens, pat, cnt, brute)
                 assert len(occurrences(tokens, pat)) >= 1
+
+
+def test_aisi_writes_pattern_artifact(tmp_path):
+    """The mined kernel-name sequence itself is exposed
+    (iteration_pattern.txt + console head)."""
+    from sofa_amd.aisi.aisi import sofa_aisi
+    from sofa_amd.config import SofaConfig
+
+    import os
+
+    df = _synth_training_trace(n_iters=20)
+    cfg = SofaConfig(logdir=str(tmp_path), num_iterations=20)
+    out = sofa_aisi(str(tmp_path), cfg, None, df, None, [])
+    assert out is not None
+    pat_path = os.path.join(str(tmp_path), "iteration_pattern.txt")
+    assert os.path.isfile(pat_path)
+    lines = open(pat_path).read().splitlines()
+    assert len(lines) >= 2
+    # names, not token ids
+    assert any("fw_" in l or "bw_" in l or "gemm" in l or "conv" in l for l in lines), lines[:5]
