@@ -117,7 +117,7 @@ def main() -> int:
         os.makedirs(logdir, exist_ok=True)
     if args.tracer == "lite" and not os.path.exists(LITE):
         args.tracer = "sdk"
-    have_tracer = os.path.exists(TRACER)
+    have_tracer = os.path.exists(TRACER) and not args.no_profile
     if have_tracer:
         setup_tracer_env(logdir, args.tracer)
         if args.tracer == "sdk":
