@@ -130,11 +130,26 @@ def elf_is_pie(path: str) -> bool:
 class DsoSymbols:
     def __init__(self, path: str):
         self.path = path
+        self._lines = None  # lazy DWARF .debug_line table (None = not tried)
         syms = read_elf_symbols(path)
         self.addrs = np.array([s[0] for s in syms], dtype=np.uint64)
         self.names = [s[2] for s in syms]
         self.sizes = np.array([s[1] for s in syms], dtype=np.uint64)
         self.pie = elf_is_pie(path)
+
+    def line_for(self, file_addr: int):
+        """(source file basename, line) via .debug_line, or None."""
+        if self._lines is None:
+            from .dwarf_lines import LineTable
+
+            self._lines = LineTable(self.path)
+        hit = self._lines.lookup(file_addr)
+        if hit is None:
+            return None
+        f, ln = hit
+        import os as _os
+
+        return _os.path.basename(f), ln
 
     def resolve(self, file_addr: int) -> Optional[str]:
         if len(self.addrs) == 0:
@@ -236,7 +251,11 @@ class Symbolizer:
                     sym = dso.resolve(file_addr)
                     base = os.path.basename(path)
                     if sym:
-                        return (demangle(sym), base)
+                        out = demangle(sym)
+                        loc = dso.line_for(file_addr)
+                        if loc is not None:
+                            out = "%s (%s:%d)" % (out, loc[0], loc[1])
+                        return (out, base)
                     return ("0x%x" % ip, base)
                 elif ip < end:
                     return ("0x%x" % ip, os.path.basename(path))

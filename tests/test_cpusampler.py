@@ -83,3 +83,58 @@ def test_demangle():
     assert demangle("_Z3addii") == "add(int, int)"
     assert demangle("main") == "main"
     assert demangle("not_mangled") == "not_mangled"
+
+
+def test_dwarf_line_table(tmp_path):
+    """DWARF .debug_line reader: file:line for a -g-compiled function."""
+    import subprocess
+
+    from sofa_amd.preprocess.dwarf_lines import LineTable
+    from sofa_amd.preprocess.symbols import read_elf_symbols
+
+    csrc = tmp_path / "hot.c"
+    csrc.write_text(
+        "int helper(int x) { return x * 3 + 1; }\n"
+        "int work(int n) {\n"
+        "  int s = 0;\n"
+        "  for (int i = 0; i < n; i++) s += helper(i);\n"
+        "  return s;\n"
+        "}\n"
+        "int main(void) { return work(10) & 0; }\n"
+    )
+    exe = tmp_path / "hot"
+    subprocess.run(
+        ["gcc", "-g", "-O0", "-no-pie", str(csrc), "-o", str(exe)],
+        check=True, capture_output=True,
+    )
+    lt = LineTable(str(exe))
+    assert lt.addrs, "no line rows parsed"
+    syms = {name: addr for addr, _, name in read_elf_symbols(str(exe))}
+    assert "work" in syms
+    hit = lt.lookup(syms["work"] + 8)
+    assert hit is not None
+    fname, line = hit
+    assert fname.endswith("hot.c")
+    assert 2 <= line <= 6, line
+
+
+def test_symbolizer_appends_file_line(tmp_path):
+    """End-to-end: a sampled IP inside a -g binary resolves to
+    'func (file.c:N)'."""
+    import subprocess
+
+    from sofa_amd.preprocess.symbols import Symbolizer, read_elf_symbols
+
+    csrc = tmp_path / "app.c"
+    csrc.write_text("int spin(int n){int s=0;for(int i=0;i<n;i++)s+=i;return s;}\n"
+                    "int main(void){return spin(5)&0;}\n")
+    exe = tmp_path / "app"
+    subprocess.run(["gcc", "-g", "-O0", "-no-pie", str(csrc), "-o", str(exe)],
+                   check=True, capture_output=True)
+    syms = {name: addr for addr, _, name in read_elf_symbols(str(exe))}
+    ip = syms["spin"] + 4
+    s = Symbolizer({42: [(0x400000, 0x100000, 0x400000, str(exe))]})
+    # non-PIE: file_addr == ip directly
+    sym, dso = s.resolve(42, ip)
+    assert sym.startswith("spin")
+    assert "app.c:" in sym, sym
