@@ -887,6 +887,12 @@ unsigned long long sofa_lite_event_count() {
   return g_n_records.load(std::memory_order_relaxed);
 }
 
+int sofa_lite_flush() {
+  std::lock_guard<std::mutex> lk(g_out_mutex);
+  if (g_out) fflush(g_out);
+  return 0;
+}
+
 int sofa_lite_active() { return g_out != nullptr; }
 
 unsigned long long sofa_lite_dropped() { return g_pool_exhausted.load(); }
