@@ -293,10 +293,13 @@ def main() -> int:
                 "measured_n_gpus": world_size,
             },
         }
-        print(json.dumps(result))
+        print(json.dumps(result), flush=True)
     if distributed:
         dist.destroy_process_group()
-    return 0
+    # skip interpreter/native teardown: the HIP/HSA unload path segfaults
+    # intermittently AFTER all output is flushed (observed with and without
+    # collectors); the result above is the contract, exit hard and clean
+    os._exit(0)
 
 
 if __name__ == "__main__":
