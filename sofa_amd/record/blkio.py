@@ -51,6 +51,18 @@ class BlkTracer(threading.Thread):
             return False
         inst = os.path.join(root, "instances", f"sofa_{os.getpid()}")
         try:
+            # reap instances from crashed/killed runs (rmdir only works when
+            # the creator is gone; EBUSY for live ones is fine)
+            inst_root = os.path.join(root, "instances")
+            if os.path.isdir(inst_root):
+                for name in os.listdir(inst_root):
+                    if name.startswith("sofa_") and name != f"sofa_{os.getpid()}":
+                        try:
+                            pid = int(name.split("_", 1)[1])
+                            if not os.path.exists(f"/proc/{pid}"):
+                                os.rmdir(os.path.join(inst_root, name))
+                        except (ValueError, OSError):
+                            pass
             os.makedirs(inst, exist_ok=True)
             with open(os.path.join(inst, "trace_clock"), "w") as f:
                 f.write("mono_raw")

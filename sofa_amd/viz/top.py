@@ -89,9 +89,12 @@ def sofa_top(interval: float = 1.0, once: bool = False) -> int:
                         busy = smi.busy_percent(dev) or 0
                         vram = (smi.memory_usage(dev) or 0) / 1e9
                         power = smi.power_watts(dev) or 0.0
+                        gm = smi.gpu_metrics(dev)
+                        mm = gm["mm_activity"] if gm else -1
+                        mm_s = (" mm %3d%%" % mm) if mm >= 0 else ""
                         print(
-                            "GPU%-2d %s  vram %6.1f GB  %5.0f W"
-                            % (dev, _bar(float(busy)), vram, power)
+                            "GPU%-2d %s  vram %6.1f GB  %5.0f W%s"
+                            % (dev, _bar(float(busy)), vram, power, mm_s)
                         )
                 elif once:
                     print("(no GPU visible)")
