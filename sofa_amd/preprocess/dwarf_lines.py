@@ -133,7 +133,12 @@ class LineTable:
     def __init__(self, path: str):
         self.addrs: List[int] = []
         self.rows: List[Tuple[str, int]] = []  # parallel to addrs
+        # cheap preflight: parse ONLY the section headers (a release
+        # libtorch is ~2 GB; slurping it to discover there is no
+        # .debug_line would double preprocess IO)
         try:
+            if not self._has_debug_line(path):
+                return
             with open(path, "rb") as f:
                 data = f.read()
         except OSError:
@@ -154,6 +159,27 @@ class LineTable:
         pairs.sort()
         self.addrs = [p[0] for p in pairs]
         self.rows = [(p[1], p[2]) for p in pairs]
+
+    @staticmethod
+    def _has_debug_line(path: str) -> bool:
+        with open(path, "rb") as f:
+            hdr = f.read(64)
+            if len(hdr) < 64 or hdr[:4] != b"\x7fELF" or hdr[4] != 2:
+                return False
+            (e_shoff,) = struct.unpack_from("<Q", hdr, 0x28)
+            (e_shentsize, e_shnum, e_shstrndx) = struct.unpack_from("<HHH", hdr, 0x3A)
+            if e_shoff == 0 or e_shnum == 0 or e_shstrndx >= e_shnum:
+                return False
+            f.seek(e_shoff)
+            sh = f.read(e_shnum * e_shentsize)
+            if len(sh) < e_shnum * e_shentsize:
+                return False
+            (str_off, str_size) = struct.unpack_from(
+                "<QQ", sh, e_shstrndx * e_shentsize + 0x18
+            )
+            f.seek(str_off)
+            shstr = f.read(str_size)
+            return b".debug_line\0" in shstr
 
     # ---- one line-number program unit
     def _parse_unit(self, r: _Reader, data, line_str, debug_str, pairs) -> None:
