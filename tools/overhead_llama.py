@@ -24,23 +24,21 @@ def main() -> int:
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--batch", type=int, default=2)
     ap.add_argument("--seq", type=int, default=2048)
+    ap.add_argument("--tracer", choices=["sdk", "lite"], default="lite")
     args = ap.parse_args()
 
     logdir = os.path.join(REPO, "gpurun_out", "llama_ovh")
     os.makedirs(logdir, exist_ok=True)
-    os.environ["SOFA_LOGDIR"] = logdir
-    os.environ["SOFA_DEFER_START"] = "1"
-    prev = os.environ.get("ROCP_TOOL_LIBRARIES", "")
-    if TRACER not in prev:
-        os.environ["ROCP_TOOL_LIBRARIES"] = TRACER + ((":" + prev) if prev else "")
+    from bench import setup_tracer_env, TracerCtl
+
+    setup_tracer_env(logdir, args.tracer)
 
     import torch
     import torch.nn.functional as F
 
     from sofa_amd.workloads.llama import build_llama8b
 
-    lib = ctypes.CDLL(TRACER)
-    lib.sofa_tracer_event_count.restype = ctypes.c_ulonglong
+    lib = TracerCtl(args.tracer)
 
     model = build_llama8b(device="cuda", n_layers=args.layers)
     opt = torch.optim.AdamW(model.parameters(), lr=1e-4)
@@ -69,18 +67,18 @@ def main() -> int:
 
     # interleaved A/B like bench.py
     t_plain = t_prof = 0.0
-    n0 = lib.sofa_tracer_event_count()
+    n0 = lib.event_count()
     per = max(args.steps // 2, 1)
     for _ in range(2):
         t_plain += timed(per)
-        lib.sofa_tracer_start()
+        lib.start()
         t_prof += timed(per)
-        lib.sofa_tracer_stop()
-    n_ev = int(lib.sofa_tracer_event_count() - n0)
+        lib.stop()
+    n_ev = int(lib.event_count() - n0)
 
     steps = per * 2
-    print("llama-%dL b%d s%d: plain %.1f ms/step, profiled %.1f ms/step, "
-          "overhead %.2f%%, %d events (%.0f ev/s)"
+    print(("llama-%dL b%d s%d [" + args.tracer + "]: plain %.1f ms/step, "
+           "profiled %.1f ms/step, overhead %.2f%%, %d events (%.0f ev/s)")
           % (args.layers, args.batch, args.seq,
              t_plain / steps * 1e3, t_prof / steps * 1e3,
              100 * (t_prof - t_plain) / t_plain, n_ev, n_ev / t_prof))

@@ -26,22 +26,21 @@ def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--launches", type=int, default=200_000)
     ap.add_argument("--streams", type=int, default=4)
+    ap.add_argument("--tracer", choices=["sdk", "lite"], default="lite")
     args = ap.parse_args()
 
     logdir = os.path.join(REPO, "gpurun_out", "stress_sgt")
     os.makedirs(logdir, exist_ok=True)
-    os.environ["SOFA_LOGDIR"] = logdir
-    os.environ["SOFA_DEFER_START"] = "1"
-    os.environ["SOFA_TRACE_HIP_API"] = "1"
-    prev = os.environ.get("ROCP_TOOL_LIBRARIES", "")
-    if TRACER not in prev:
-        os.environ["ROCP_TOOL_LIBRARIES"] = TRACER + ((":" + prev) if prev else "")
+    from bench import setup_tracer_env, TracerCtl
+
+    setup_tracer_env(logdir, args.tracer)
+    if args.tracer == "sdk":
+        os.environ["SOFA_TRACE_HIP_API"] = "1"
 
     import torch
 
     assert torch.cuda.is_available()
-    lib = ctypes.CDLL(TRACER)
-    lib.sofa_tracer_event_count.restype = ctypes.c_ulonglong
+    lib = TracerCtl(args.tracer)
 
     streams = [torch.cuda.Stream() for _ in range(args.streams)]
     x = [torch.zeros(64, device="cuda") for _ in range(args.streams)]
@@ -62,14 +61,14 @@ def main() -> int:
     plain_rate = args.launches / plain_s
 
     # traced rate
-    lib.sofa_tracer_start()
+    lib.start()
     storm(2000)
-    n0 = lib.sofa_tracer_event_count()
+    n0 = lib.event_count()
     t0 = time.perf_counter()
     storm(args.launches)
     traced_s = time.perf_counter() - t0
-    lib.sofa_tracer_stop()
-    n_events = int(lib.sofa_tracer_event_count() - n0)
+    lib.stop()
+    n_events = int(lib.event_count() - n0)
     traced_rate = args.launches / traced_s
 
     from sofa_amd.preprocess.sgt import parse_sgt
