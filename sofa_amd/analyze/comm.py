@@ -189,8 +189,9 @@ def attach_kernel_times(
         order = np.argsort(k_ts, kind="stable")
         k_ts, k_dur = k_ts[order], k_dur[order]
         # RCCL debug-log rows carry no wall-clock timestamps (lite mode,
-        # preprocess.rccl_log) — pure order matching, no time constraint
-        order_only = bool(np.all(api_ts[sel] < 1.0))
+        # preprocess.rccl_log writes order*1e-9, i.e. sub-millisecond) —
+        # pure order matching, no time constraint
+        order_only = bool(np.all(api_ts[sel] < 1e-3))
         ki = 0
         for idx in sel:
             # first unconsumed kernel starting at/after the API start
