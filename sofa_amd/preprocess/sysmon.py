@@ -230,6 +230,16 @@ def parse_gpusmi(logdir: str, tb: Optional[TimeBase]):
         return new_trace_df(0), new_trace_df(0), None
     if "mm" not in raw.columns:
         raw["mm"] = -1
+    # ragged/corrupt rows: coerce everything numeric, drop rows missing the
+    # required fields, default the optional ones (fuzz-hardened)
+    for c in raw.columns:
+        raw[c] = pd.to_numeric(raw[c], errors="coerce")
+    raw = raw.dropna(subset=["ts", "dev", "busy"])
+    raw["membusy"] = raw["membusy"].fillna(-1)
+    raw["vram"] = raw["vram"].fillna(-1)
+    raw["power"] = raw["power"].fillna(-1.0)
+    raw["mm"] = raw["mm"].fillna(-1)
+    raw["dev"] = raw["dev"].astype(np.int64)
     d = raw[raw["busy"] >= 0]
     if len(d) == 0:
         return new_trace_df(0), new_trace_df(0), None
