@@ -60,6 +60,18 @@ def write_chrome_trace(logdir: str, pre: Optional[dict] = None, path: str = "chr
     if df_gpu is not None:
         kernels = df_gpu[df_gpu["copyKind"] == 0]
         copies = df_gpu[df_gpu["copyKind"] != 0]
+        # device-ring software events get their own track group, laned by
+        # producer id (the `src` a kernel passed to ring_push, e.g.
+        # workgroup index) instead of being mixed into the kernel lanes
+        names = kernels["name"].astype(str)
+        devring = kernels[names.str.contains("devring:", regex=False)]
+        kernels = kernels[~names.str.contains("devring:", regex=False)]
+        if len(devring):
+            devring = devring.copy()
+            # lane = ring_push's producer id (ring_writer maps RingRec.src
+            # into tid), mod 64 to bound the track count
+            devring["deviceId"] = devring["tid"].astype(int) % 64
+            events += _events_from_trace(devring, "devring", "deviceId", "devring")
         events += _events_from_trace(kernels, "GPU kernels", "deviceId", "kernel")
         events += _events_from_trace(copies, "GPU copies", "deviceId", "memcpy")
     df_rccl = get("df_rccl", "rccltrace.csv")
