@@ -838,6 +838,29 @@ void finalize() {
   if (!done.compare_exchange_strong(expected, true)) return;
   g_shutdown.store(true, std::memory_order_release);
   if (g_reaper_started.load()) pthread_join(g_reaper, nullptr);
+  // destroy every pooled signal BEFORE the runtime shuts down: leaked
+  // GPU-only signals crash ROCr teardown (measured: lite-full exits 139,
+  // lite-prof/off exit 0 — the only delta is the signal pools)
+  {
+    std::lock_guard<std::mutex> lk(g_queues_mutex);
+    for (auto* q : g_queues) {
+      for (auto& s : q->slots) {
+        if (s.sig.handle) {
+          g_core.hsa_signal_destroy_fn(s.sig);
+          s.sig.handle = 0;
+        }
+      }
+    }
+  }
+  {
+    std::lock_guard<std::mutex> lk(g_copy_mutex);
+    for (auto* s : g_copy_free) {
+      if (s->sig.handle) {
+        g_core.hsa_signal_destroy_fn(s->sig);
+        s->sig.handle = 0;
+      }
+    }
+  }
   if (env_flag("SOFA_LITE_DEBUG", false)) {
     fprintf(stderr,
             "[sofahsalite] queues=%lu batches=%lu kernel_pkts=%lu attached=%lu "
