@@ -294,6 +294,12 @@ def parse_xgmi_counters(logdir: str, tb: Optional[TimeBase]):
     raw = _read(logdir, "xgmi_counters.txt", cols)
     if raw is None or len(raw) < 2:
         return new_trace_df(0), None
+    for c in raw.columns:
+        raw[c] = pd.to_numeric(raw[c], errors="coerce")
+    raw = raw.dropna(subset=["ts", "dev"]).fillna(0)
+    if len(raw) < 2:
+        return new_trace_df(0), None
+    raw["dev"] = raw["dev"].astype(np.int64)
     frames = []
     csv_rows = []
     for dev, grp in raw.groupby("dev"):
