@@ -81,6 +81,11 @@ std::mutex g_out_mutex;
 std::atomic<uint64_t> g_n_records{0};
 std::atomic<bool> g_armed{true};
 std::atomic<bool> g_shutdown{false};
+// set once the REAL hsa_shut_down has run: our cached table pointers skip
+// ROCr's public liveness guards, so every wrapper must refuse calls on a
+// dead runtime (ROCclr's Device::~Device destroys queues after shutdown —
+// measured: that tail call segfaulted)
+std::atomic<bool> g_runtime_down{false};
 
 enum LiteMode { MODE_OFF = 0, MODE_PROF = 1, MODE_FULL = 2 };
 LiteMode g_mode = MODE_FULL;
@@ -500,6 +505,8 @@ hsa_status_t queue_create_wrap(hsa_agent_t agent, uint32_t size,
                                void (*callback)(hsa_status_t, hsa_queue_t*, void*),
                                void* data, uint32_t private_segment_size,
                                uint32_t group_segment_size, hsa_queue_t** queue) {
+  if (g_runtime_down.load(std::memory_order_acquire))
+    return HSA_STATUS_ERROR_NOT_INITIALIZED;
   enumerate_agents();
   hsa_device_type_t dev_type = HSA_DEVICE_TYPE_CPU;
   g_core.hsa_agent_get_info_fn(agent, HSA_AGENT_INFO_DEVICE, &dev_type);
@@ -561,10 +568,13 @@ hsa_status_t hsa_shut_down_wrap() {
   // stop the reaper and close the file while signals are still alive; ROCr
   // refcounts init/shutdown but one shutdown from the app is the exit path
   finalize_for_shutdown();
+  g_runtime_down.store(true, std::memory_order_release);
   return g_core.hsa_shut_down_fn();
 }
 
 hsa_status_t queue_destroy_wrap(hsa_queue_t* queue) {
+  if (g_runtime_down.load(std::memory_order_acquire))
+    return HSA_STATUS_ERROR_NOT_INITIALIZED;
   // drain: reaper keeps polling; just mark dead and let slots finish.
   // ROCr destroys the proxy after in-flight packets retire, so pending
   // signals have fired by then; a short grace wait covers the reap gap.
@@ -593,6 +603,8 @@ hsa_status_t queue_destroy_wrap(hsa_queue_t* queue) {
 // ---- kernel symbol names: executable freeze -> NameRec(kernel_object)
 
 hsa_status_t exe_freeze_wrap(hsa_executable_t executable, const char* options) {
+  if (g_runtime_down.load(std::memory_order_acquire))
+    return HSA_STATUS_ERROR_NOT_INITIALIZED;
   hsa_status_t st = g_core.hsa_executable_freeze_fn(executable, options);
   if (st != HSA_STATUS_SUCCESS || !g_out) return st;
   g_core.hsa_executable_iterate_symbols_fn(
@@ -707,6 +719,8 @@ hsa_status_t async_copy_wrap(void* dst, hsa_agent_t dst_agent, const void* src,
                              uint32_t num_dep_signals,
                              const hsa_signal_t* dep_signals,
                              hsa_signal_t completion_signal) {
+  if (g_runtime_down.load(std::memory_order_acquire))
+    return HSA_STATUS_ERROR_NOT_INITIALIZED;
   g_stats.copy_calls.fetch_add(1, std::memory_order_relaxed);
   // ROCclr passes completion_signal.handle == 0 on its SDMA path (measured
   // on MI355X); a zero signal means nobody waits on it, so attaching ours is
@@ -752,6 +766,8 @@ hsa_status_t async_copy_engine_wrap(void* dst, hsa_agent_t dst_agent,
                                     hsa_signal_t completion_signal,
                                     hsa_amd_sdma_engine_id_t engine_id,
                                     bool force_copy_on_sdma) {
+  if (g_runtime_down.load(std::memory_order_acquire))
+    return HSA_STATUS_ERROR_NOT_INITIALIZED;
   g_stats.copy_engine_calls.fetch_add(1, std::memory_order_relaxed);
   if (g_mode != MODE_FULL || !g_armed.load(std::memory_order_relaxed)) {
     return g_amd.hsa_amd_memory_async_copy_on_engine_fn(
@@ -795,6 +811,8 @@ hsa_status_t async_copy_rect_wrap(const hsa_pitched_ptr_t* dst,
                                   uint32_t num_dep_signals,
                                   const hsa_signal_t* dep_signals,
                                   hsa_signal_t completion_signal) {
+  if (g_runtime_down.load(std::memory_order_acquire))
+    return HSA_STATUS_ERROR_NOT_INITIALIZED;
   g_stats.copy_rect_calls.fetch_add(1, std::memory_order_relaxed);
   return g_amd.hsa_amd_memory_async_copy_rect_fn(
       dst, dst_offset, src, src_offset, range, copy_agent, dir,
