@@ -332,7 +332,8 @@ void drain_submit_bufs(std::vector<char>& chunk) {
 void on_submit(const void* pkts, uint64_t pkt_count, uint64_t /*user_pkt_index*/,
                void* data, hsa_amd_queue_intercept_packet_writer writer) {
   auto* q = static_cast<QueueCtx*>(data);
-  if (g_mode != MODE_FULL || !g_armed.load(std::memory_order_relaxed)) {
+  if (g_mode != MODE_FULL || !g_armed.load(std::memory_order_relaxed) ||
+      g_shutdown.load(std::memory_order_relaxed)) {
     writer(pkts, pkt_count);
     return;
   }
