@@ -95,6 +95,7 @@ def build_all(verbose: bool = False) -> None:
             [
                 "g++",
                 "-O2",
+                "-g",  # line info for crash triage; free at runtime
                 "-std=c++17",
                 "-fPIC",
                 "-shared",

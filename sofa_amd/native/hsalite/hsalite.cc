@@ -568,12 +568,17 @@ void finalize_for_shutdown();
 hsa_status_t hsa_shut_down_wrap() {
   // stop the reaper and close the file while signals are still alive; ROCr
   // refcounts init/shutdown but one shutdown from the app is the exit path
+  if (env_flag("SOFA_LITE_DEBUG", false))
+    fprintf(stderr, "[sofahsalite] hsa_shut_down intercepted\n");
   finalize_for_shutdown();
   g_runtime_down.store(true, std::memory_order_release);
   return g_core.hsa_shut_down_fn();
 }
 
 hsa_status_t queue_destroy_wrap(hsa_queue_t* queue) {
+  if (env_flag("SOFA_LITE_DEBUG", false))
+    fprintf(stderr, "[sofahsalite] queue_destroy %p (down=%d)\n",
+            (void*) queue, (int) g_runtime_down.load());
   if (g_runtime_down.load(std::memory_order_acquire))
     return HSA_STATUS_ERROR_NOT_INITIALIZED;
   // drain: reaper keeps polling; just mark dead and let slots finish.
