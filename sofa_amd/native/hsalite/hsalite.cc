@@ -161,7 +161,14 @@ struct AgentInfo {
   int32_t device = -1;  // GPU index in HSA enumeration order; -1 for CPU
   hsa_agent_t agent{};
 };
-std::unordered_map<uint64_t, AgentInfo> g_agents;
+// LEAKED on purpose: this library's static destructors run before
+// ROCclr's RuntimeTearDown (we are dlopened after libamdhip64), and
+// ROCclr still calls our queue wrappers from its teardown — iterating a
+// destructed container was the round-2 exit crash (gdb: q=0x555000e3bb91).
+std::unordered_map<uint64_t, AgentInfo>& agents_map() {
+  static auto* m = new std::unordered_map<uint64_t, AgentInfo>();
+  return *m;
+}
 std::mutex g_agents_mutex;
 std::atomic<bool> g_agents_done{false};
 
@@ -201,7 +208,7 @@ void enumerate_agents() {
           rec.name[sizeof(rec.name) - 1] = 0;
           write_raw(&rec, sizeof(rec));
         }
-        g_agents[agent.handle] = info;
+        agents_map()[agent.handle] = info;
         return HSA_STATUS_SUCCESS;
       },
       &ctx);
@@ -209,8 +216,9 @@ void enumerate_agents() {
 }
 
 int32_t agent_device(hsa_agent_t a) {
-  auto it = g_agents.find(a.handle);
-  return it == g_agents.end() ? -1 : it->second.device;
+  auto& m = agents_map();
+  auto it = m.find(a.handle);
+  return it == m.end() ? -1 : it->second.device;
 }
 
 // ------------------------------------------------------- per-queue dispatch
@@ -244,7 +252,10 @@ struct QueueCtx {
 };
 
 std::mutex g_queues_mutex;
-std::vector<QueueCtx*> g_queues;  // never shrunk; ctx leak on destroy is fine
+std::vector<QueueCtx*>& queues_vec() {  // leaked (see agents_map comment)
+  static auto* v = new std::vector<QueueCtx*>();
+  return *v;
+}
 
 constexpr uint32_t kSlotsPerQueue = 8192;
 
@@ -297,13 +308,16 @@ struct SubmitBuf {
   }
 };
 std::mutex g_subufs_mutex;
-std::vector<SubmitBuf*> g_subufs;
+std::vector<SubmitBuf*>& subufs_vec() {  // leaked (see agents_map comment)
+  static auto* v = new std::vector<SubmitBuf*>();
+  return *v;
+}
 
 void submit_buf_push(const sgt::ApiRec& rec) {
   static thread_local SubmitBuf* buf = [] {
     auto* b = new SubmitBuf();
     std::lock_guard<std::mutex> lk(g_subufs_mutex);
-    g_subufs.push_back(b);
+    subufs_vec().push_back(b);
     return b;
   }();
   pthread_spin_lock(&buf->lock);
@@ -315,7 +329,7 @@ void drain_submit_bufs(std::vector<char>& chunk) {
   std::vector<SubmitBuf*> bufs;
   {
     std::lock_guard<std::mutex> lk(g_subufs_mutex);
-    bufs = g_subufs;
+    bufs = subufs_vec();
   }
   for (auto* b : bufs) {
     pthread_spin_lock(&b->lock);
@@ -417,13 +431,13 @@ void* reaper_main(void*) {
     size_t nq = 0;
     {
       std::lock_guard<std::mutex> lk(g_queues_mutex);
-      nq = g_queues.size();
+      nq = queues_vec().size();
     }
     for (size_t qi = 0; qi < nq; ++qi) {
       QueueCtx* q;
       {
         std::lock_guard<std::mutex> lk(g_queues_mutex);
-        q = g_queues[qi];
+        q = queues_vec()[qi];
       }
       // swap out the inflight list
       pthread_spin_lock(&q->lock);
@@ -553,7 +567,7 @@ hsa_status_t queue_create_wrap(hsa_agent_t agent, uint32_t size,
       g_amd.hsa_amd_queue_intercept_register_fn(*queue, on_submit, q);
       {
         std::lock_guard<std::mutex> lk(g_queues_mutex);
-        g_queues.push_back(q);
+        queues_vec().push_back(q);
       }
       ensure_reaper();
     } else {
@@ -586,14 +600,14 @@ hsa_status_t queue_destroy_wrap(hsa_queue_t* queue) {
   // signals have fired by then; a short grace wait covers the reap gap.
   {
     std::lock_guard<std::mutex> lk(g_queues_mutex);
-    for (auto* q : g_queues)
+    for (auto* q : queues_vec())
       if (q->queue == queue) q->alive.store(false);
   }
   for (int spin = 0; spin < 50; ++spin) {
     bool pending = false;
     {
       std::lock_guard<std::mutex> lk(g_queues_mutex);
-      for (auto* q : g_queues)
+      for (auto* q : queues_vec())
         if (q->queue == queue) {
           pthread_spin_lock(&q->lock);
           pending = !q->inflight.empty();
@@ -652,7 +666,10 @@ struct CopySlot {
 };
 
 std::mutex g_copy_mutex;
-std::vector<CopySlot*> g_copy_free;
+std::vector<CopySlot*>& copy_free_vec() {  // leaked (see agents_map comment)
+  static auto* v = new std::vector<CopySlot*>();
+  return *v;
+}
 std::atomic<bool> g_copy_prof_enabled{false};
 
 // hsa_amd_profiling_async_copy_enable cannot run from OnLoad (mid-hsa_init);
@@ -668,9 +685,10 @@ void ensure_copy_profiling() {
 
 CopySlot* copy_slot_get() {
   std::lock_guard<std::mutex> lk(g_copy_mutex);
-  if (!g_copy_free.empty()) {
-    CopySlot* s = g_copy_free.back();
-    g_copy_free.pop_back();
+  auto& v = copy_free_vec();
+  if (!v.empty()) {
+    CopySlot* s = v.back();
+    v.pop_back();
     return s;
   }
   auto* s = new CopySlot();
@@ -684,7 +702,7 @@ CopySlot* copy_slot_get() {
 void copy_slot_put(CopySlot* s) {
   g_core.hsa_signal_store_screlease_fn(s->sig, 1);
   std::lock_guard<std::mutex> lk(g_copy_mutex);
-  g_copy_free.push_back(s);
+  copy_free_vec().push_back(s);
 }
 
 std::atomic<uint64_t> g_copy_handler_fired{0};
@@ -867,7 +885,7 @@ void finalize() {
   // lite-prof/off exit 0 — the only delta is the signal pools)
   {
     std::lock_guard<std::mutex> lk(g_queues_mutex);
-    for (auto* q : g_queues) {
+    for (auto* q : queues_vec()) {
       for (auto& s : q->slots) {
         if (s.sig.handle) {
           g_core.hsa_signal_destroy_fn(s.sig);
@@ -878,7 +896,7 @@ void finalize() {
   }
   {
     std::lock_guard<std::mutex> lk(g_copy_mutex);
-    for (auto* s : g_copy_free) {
+    for (auto* s : copy_free_vec()) {
       if (s->sig.handle) {
         g_core.hsa_signal_destroy_fn(s->sig);
         s->sig.handle = 0;
