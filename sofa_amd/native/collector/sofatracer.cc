@@ -61,8 +61,13 @@ FILE* g_out = nullptr;
 std::mutex g_mutex;
 std::atomic<uint64_t> g_n_records{0};
 
-// agent handle -> logical GPU index (or -1 for CPU agents)
-std::unordered_map<uint64_t, int32_t> g_agent_device;
+// agent handle -> logical GPU index (or -1 for CPU agents).  Leaked on
+// purpose: tool-library statics are destructed before the runtimes stop
+// calling back (same hazard fixed in hsalite.cc).
+std::unordered_map<uint64_t, int32_t>& agent_device_map() {
+  static auto* m = new std::unordered_map<uint64_t, int32_t>();
+  return *m;
+}
 
 bool env_flag(const char* name, bool dflt) {
   const char* v = getenv(name);
@@ -127,8 +132,9 @@ void write_clock_rec() {
 }
 
 int32_t agent_device(rocprofiler_agent_id_t id) {
-  auto it = g_agent_device.find(id.handle);
-  return it == g_agent_device.end() ? -1 : it->second;
+  auto& m = agent_device_map();
+  auto it = m.find(id.handle);
+  return it == m.end() ? -1 : it->second;
 }
 
 // RCCL datatype element sizes (ncclDataType_t order: int8, uint8, int32,
@@ -431,7 +437,7 @@ void write_agents() {
           int32_t dev = (a->type == ROCPROFILER_AGENT_TYPE_GPU)
                             ? a->logical_node_type_id
                             : -1;
-          g_agent_device[a->id.handle] = dev;
+          agent_device_map()[a->id.handle] = dev;
           sgt::AgentRec rec{};
           rec.h = {sgt::REC_AGENT, sizeof(sgt::AgentRec), 0};
           rec.agent_handle = a->id.handle;
