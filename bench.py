@@ -113,8 +113,9 @@ def main() -> int:
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
 
     logdir = os.path.join(REPO, "gpurun_out", "bench_sgt")
-    if rank == 0:
-        os.makedirs(logdir, exist_ok=True)
+    # EVERY rank creates it: the collector opens its output during HIP init,
+    # which can precede rank 0's mkdir in a simultaneous torchrun launch
+    os.makedirs(logdir, exist_ok=True)
     if args.tracer == "lite" and not os.path.exists(LITE):
         args.tracer = "sdk"
     have_tracer = os.path.exists(TRACER) and not args.no_profile
