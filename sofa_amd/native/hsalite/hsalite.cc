@@ -367,7 +367,10 @@ void on_submit(const void* pkts, uint64_t pkt_count, uint64_t /*user_pkt_index*/
       continue;
     }
 
-    // acquire a slot + signal
+    // acquire a slot + publish to the in-flight list in ONE critical
+    // section (an unpublished-packet slot is harmless to the reaper: its
+    // signal still reads 1, so it just stays pending) — halves the lock
+    // traffic on the hot submit path
     pthread_spin_lock(&q->lock);
     if (q->free_idx.empty()) {
       pthread_spin_unlock(&q->lock);
@@ -376,6 +379,7 @@ void on_submit(const void* pkts, uint64_t pkt_count, uint64_t /*user_pkt_index*/
     }
     uint32_t idx = q->free_idx.back();
     q->free_idx.pop_back();
+    q->inflight.push_back(idx);
     pthread_spin_unlock(&q->lock);
 
     PendSlot& s = q->slots[idx];
@@ -396,10 +400,6 @@ void on_submit(const void* pkts, uint64_t pkt_count, uint64_t /*user_pkt_index*/
       patched = true;
     }
     scratch[i].completion_signal = s.sig;
-
-    pthread_spin_lock(&q->lock);
-    q->inflight.push_back(idx);
-    pthread_spin_unlock(&q->lock);
     g_stats.attached.fetch_add(1, std::memory_order_relaxed);
   }
   writer(patched ? (const void*) scratch.data() : pkts, pkt_count);
