@@ -91,6 +91,17 @@ def advise(f: Dict[str, float]) -> List[Hint]:
                 "xlink_traffic.csv for the hot link",
             )
         )
+    if g("gpu_underfill_time_ratio", 0.0) > 0.25:
+        hints.append(
+            (
+                "gpu_underfill_time_ratio",
+                "%.0f%% of kernel time is in launches with <256 workgroups"
+                % (g("gpu_underfill_time_ratio") * 100),
+                "MI355X has 256 CUs over 8 XCDs: batch more work per launch "
+                "(bigger batch/fused ops) or capture the launch-bound loop "
+                "in a hipGraph; per-kernel tuning cannot fill the chip here",
+            )
+        )
     if g("rccl_hot_link_bytes", 0.0) > 0:
         hints.append(
             (

@@ -90,6 +90,28 @@ def gpu_profile(df_gpu: pd.DataFrame, df_rccl: pd.DataFrame, features: Features)
             short = str(name) if len(str(name)) < 90 else str(name)[:87] + "..."
             print("    %8.4f s %6d  %s" % (row["sum"], int(row["count"]), short))
 
+    # chip-underfill: MI355X = 256 CUs across 8 XCDs; launches with fewer
+    # than 256 workgroups cannot fill the chip (payload carries the WG count
+    # for kernel rows — preprocess.gpu).  Report the GPU-time share spent in
+    # such launches; a high share means small-kernel/launch-shape problems
+    # that no amount of per-kernel tuning fixes.
+    kt = kernels["duration"].sum()
+    if kt > 0 and (kernels["payload"] > 0).any():
+        under = kernels[(kernels["payload"] > 0) & (kernels["payload"] < 256)]
+        ratio = float(under["duration"].sum() / kt)
+        features.append(("gpu_underfill_time_ratio", ratio))
+        if ratio > 0.01:
+            print(
+                "  chip underfill: %.1f%% of kernel time in launches with <256 "
+                "workgroups (256 CUs / 8 XCDs need >=256 WGs to fill)" % (ratio * 100)
+            )
+            worst = (
+                under.groupby("name")["duration"].sum().sort_values(ascending=False).head(3)
+            )
+            for name, dur in worst.items():
+                short = str(name) if len(str(name)) < 80 else str(name)[:77] + "..."
+                print("    %8.4f s  %s" % (dur, short))
+
 
 def gpusmi_profile(df_sm: pd.DataFrame, features: Features, logdir: str = "") -> None:
     if df_sm is None or len(df_sm) == 0:

@@ -126,11 +126,24 @@ def sgt_to_gputrace(
                 ],
                 dtype=object,
             )
+            # payload (unused for kernels in the reference schema) carries
+            # the WORKGROUP count: MI355X has 256 CUs over 8 XCDs, so
+            # launches with <256 workgroups underfill the chip — analyze
+            # surfaces the time share of such launches (occupancy hint)
+            wg = np.maximum(k["wg_x"].astype(np.int64), 1) * np.maximum(
+                k["wg_y"].astype(np.int64), 1
+            ) * np.maximum(k["wg_z"].astype(np.int64), 1)
+            grid = k["grid_x"].astype(np.int64) * np.maximum(
+                k["grid_y"].astype(np.int64), 1
+            ) * np.maximum(k["grid_z"].astype(np.int64), 1)
+            n_wgs = np.maximum(grid // np.maximum(wg, 1), 1)
             emit(
                 len(k),
                 timestamp=_timeline(tb, sgt, k["start_ns"]),
                 duration=(k["end_ns"] - k["start_ns"]).astype(np.float64) * 1e-9,
                 deviceId=k["device"].astype(np.int64),
+                payload=n_wgs,
+                pkt_src=k["group_segment_size"].astype(np.int64),  # LDS bytes
                 pid=np.full(len(k), sgt.pid, dtype=np.int64),
                 tid=k["tid"].astype(np.int64),
                 event=k["kernel_id"].astype(np.float64),

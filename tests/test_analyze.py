@@ -308,3 +308,25 @@ def test_rccl_log_order_matched_to_kernels(tmp_path):
     dur, matched = comm_mod.attach_kernel_times(df_rccl, kern)
     assert matched.all()
     assert np.allclose(sorted(dur), [2e-3, 3e-3, 4e-3])
+
+
+def test_gpu_underfill_detection(capsys):
+    """MI355X chip-underfill: GPU-time share of launches with <256
+    workgroups (payload column carries WG counts for kernel rows)."""
+    df = new_trace_df(4)
+    df["copyKind"] = 0
+    df["name"] = ["big_kernel", "big_kernel", "tiny_kernel", "tiny_kernel"]
+    df["payload"] = [4096, 4096, 8, 8]  # workgroups
+    df["duration"] = [1e-3, 1e-3, 3e-3, 3e-3]  # tiny kernels dominate time
+    feats = []
+    profiles.gpu_profile(df, None, feats)
+    d = dict(feats)
+    assert abs(d["gpu_underfill_time_ratio"] - 0.75) < 1e-9
+    out = capsys.readouterr().out
+    assert "chip underfill" in out
+    assert "tiny_kernel" in out
+
+    from sofa_amd.advisor.rules import advise
+
+    hints = advise(d)
+    assert any("256 CUs" in h[2] for h in hints)
