@@ -16,8 +16,9 @@ from typing import List, Optional
 # The unified trace schema (reference bin/sofa_config.py:49-62).  Semantics are
 # source-dependent: for CPU samples `event` = log10(ip) and `duration` =
 # period/MHz; for GPU rows `copyKind` encodes the copy class (0 kernel, 1 H2D,
-# 2 D2H, 8 D2D, 10 P2P, 16 RCCL collective); for network rows pkt_src/pkt_dst
-# are IPv4 addresses packed base-1000.
+# 2 D2H, 8 D2D, 10 P2P, 16 RCCL collective), kernel rows carry the workgroup
+# count in `payload` and LDS bytes in `pkt_src` (chip-underfill analysis);
+# for network rows pkt_src/pkt_dst are IPv4 addresses packed base-1000.
 TRACE_COLUMNS = [
     "timestamp",
     "event",
