@@ -266,7 +266,7 @@ bool g_submit_spans = true;  // SOFA_LITE_SUBMIT_SPANS=0 to disable
 
 std::atomic<uint64_t> g_pool_exhausted{0};
 
-// diagnostics (SOFA_LITE_DEBUG=1 prints at exit; sofa_lite_stats exports)
+// diagnostics (SOFA_LITE_DEBUG=1 prints the full set at exit)
 struct Stats {
   std::atomic<uint64_t> queue_create_gpu{0};
   std::atomic<uint64_t> submit_batches{0};
