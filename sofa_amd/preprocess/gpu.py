@@ -110,19 +110,25 @@ def sgt_to_gputrace(
                 resolved = {kid: demangle(nm) for kid, nm in names.items()}
             else:
                 resolved = names
-            # vectorized naming: format only unique (device, kernel_id) pairs.
-            # Structured key, not bit-packing: the lite collector's kernel_id
-            # is the 64-bit HSA kernel_object handle, which does not fit in
-            # 32 bits.
-            key = np.empty(len(k), dtype=[("dev", "<i8"), ("kid", "<u8")])
-            key["dev"] = k["device"]
-            key["kid"] = k["kernel_id"]
-            uniq, inv = np.unique(key, return_inverse=True)
+            # vectorized naming: format only unique (device, kernel_id)
+            # pairs.  Factorized integer codes, not bit-packing (the lite
+            # collector's kernel_id is a full 64-bit kernel_object handle)
+            # and not a structured dtype (void-compare np.unique measured
+            # 2x slower end-to-end at 1M events).
+            kid = k["kernel_id"].astype(np.uint64)
+            dev = k["device"].astype(np.int64)
+            inv_k, uk = pd.factorize(kid)      # hash-based, no sort
+            inv_d, ud = pd.factorize(dev)
+            code = inv_d.astype(np.int64) * len(uk) + inv_k
+            inv, uc = pd.factorize(code)
             uniq_names = np.array(
                 [
                     "[gpu%d] %s"
-                    % (u["dev"], resolved.get(int(u["kid"]), "kernel_%d" % u["kid"]))
-                    for u in uniq
+                    % (
+                        ud[c // len(uk)],
+                        resolved.get(int(uk[c % len(uk)]), "kernel_%d" % uk[c % len(uk)]),
+                    )
+                    for c in uc
                 ],
                 dtype=object,
             )
