@@ -17,6 +17,7 @@ setup(
             "sofaboard/*",
             "native/*/*.cc",
             "native/*/*.hip",
+            "native/*/*.h",
             "native/collector/*.h",
             "pystacks_inject/*.py",
         ]
