@@ -138,3 +138,63 @@ def test_symbolizer_appends_file_line(tmp_path):
     sym, dso = s.resolve(42, ip)
     assert sym.startswith("spin")
     assert "app.c:" in sym, sym
+
+
+def test_fold_stacks_synthetic(tmp_path):
+    """flame.fold_stacks on synthetic -g callchain samples: root-first
+    folding with counts (flamegraph.folded contract)."""
+    import struct
+
+    import numpy as np
+
+    from sofa_amd.preprocess.flame import fold_stacks, write_folded
+    from sofa_amd.preprocess.scs import MAX_FRAMES, SAMPLE_CS_DTYPE, ScsFile
+
+    scs = ScsFile()
+    n = 6
+    cs = np.zeros(n, dtype=SAMPLE_CS_DTYPE)
+    cs["type"] = 6
+    cs["size"] = SAMPLE_CS_DTYPE.itemsize
+    cs["pid"] = 42
+    cs["tid"] = 42
+    cs["n_frames"] = 2
+    for i in range(n):
+        # leaf-first frames: leaf=0x2000+ variation, root=0x1000
+        cs["frames"][i][0] = 0x2000 + (0x10 if i >= 4 else 0)
+        cs["frames"][i][1] = 0x1000
+    scs.samples_cs = cs
+    scs.comms = {42: "worker"}
+    out = fold_stacks(scs)
+    assert sum(out.values()) == n
+    # two distinct stacks: 4 samples of one, 2 of the other; root first
+    counts = sorted(out.values())
+    assert counts == [2, 4]
+    for key in out:
+        parts = key.split(";")
+        assert parts[0] == "worker"  # comm root
+    path = write_folded(scs, str(tmp_path))
+    assert path and os.path.isfile(path)
+    lines = open(path).read().splitlines()
+    assert len(lines) == 2 and all(l.rsplit(" ", 1)[1].isdigit() for l in lines)
+
+
+def test_monitor_main_subprocess(tmp_path):
+    """record.monitor_main as the bench uses it: writes poller files and
+    exits on SIGTERM."""
+    import signal
+    import subprocess
+    import sys
+    import time
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "sofa_amd.record.monitor_main",
+         "--logdir", str(tmp_path), "--rate", "20", "--no-gpu"],
+        cwd=repo,
+    )
+    time.sleep(1.5)
+    proc.send_signal(signal.SIGTERM)
+    assert proc.wait(timeout=10) == 0
+    mp = os.path.join(str(tmp_path), "mpstat.txt")
+    assert os.path.isfile(mp) and os.path.getsize(mp) > 0
+    assert os.path.isfile(os.path.join(str(tmp_path), "vmstat.txt"))
