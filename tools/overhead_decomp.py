@@ -70,6 +70,7 @@ def configs(logdir):
         ("lite-prof", {**base, "HSA_TOOLS_LIB": lite, "SOFA_LITE_MODE": "prof"}),
         ("lite-full", {**base, "HSA_TOOLS_LIB": lite}),
         ("lite-rs", {**base, "HSA_TOOLS_LIB": lite, "SOFA_LITE_REPLACE_SIGNALS": "1"}),
+        ("lite-nospans", {**base, "HSA_TOOLS_LIB": lite, "SOFA_LITE_SUBMIT_SPANS": "0"}),
         ("sdk-null", {**base, "ROCP_TOOL_LIBRARIES": sdk, "SOFA_NULL_SINK": "1",
                       "SOFA_TRACE_HIP_API": "1", "SOFA_TRACE_RCCL": "1"}),
         ("sdk-full", {**base, "ROCP_TOOL_LIBRARIES": sdk,
