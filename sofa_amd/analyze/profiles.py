@@ -90,6 +90,34 @@ def gpu_profile(df_gpu: pd.DataFrame, df_rccl: pd.DataFrame, features: Features)
             short = str(name) if len(str(name)) < 90 else str(name)[:87] + "..."
             print("    %8.4f s %6d  %s" % (row["sum"], int(row["count"]), short))
 
+    # achieved kernel concurrency per device: sum(kernel durations) over the
+    # union of their busy intervals.  1.0 = fully serialized streams; >1
+    # means overlapping streams are actually overlapping on-device.  Pairs
+    # with the launch-gap analysis (launch.py) to separate "launch-bound"
+    # from "stream-serialized".
+    for dev, grp in kernels.groupby("deviceId"):
+        if len(grp) < 2:
+            continue
+        t0 = grp["timestamp"].to_numpy(dtype=np.float64)
+        t1 = t0 + grp["duration"].to_numpy(dtype=np.float64)
+        order = np.argsort(t0, kind="stable")
+        t0, t1 = t0[order], t1[order]
+        # merged-interval union, vectorized: a new busy segment starts where
+        # this kernel begins after the running max of previous ends
+        run_end = np.maximum.accumulate(t1)
+        gap_starts = t0[1:] > run_end[:-1]
+        seg_start = np.concatenate(([t0[0]], t0[1:][gap_starts]))
+        seg_end = np.concatenate((run_end[:-1][gap_starts], [run_end[-1]]))
+        union = float(np.maximum(seg_end - seg_start, 0).sum())
+        total = float(grp["duration"].sum())
+        if union > 0:
+            cf = total / union
+            features.append((f"gpu{int(dev)}_concurrency_factor", float(cf)))
+            print(
+                "  gpu%-2d kernel concurrency: %.2fx (busy %.4f s, serialized "
+                "sum %.4f s)" % (dev, cf, union, total)
+            )
+
     # chip-underfill: MI355X = 256 CUs across 8 XCDs; launches with fewer
     # than 256 workgroups cannot fill the chip (payload carries the WG count
     # for kernel rows — preprocess.gpu).  Report the GPU-time share spent in

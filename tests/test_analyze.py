@@ -330,3 +330,31 @@ def test_gpu_underfill_detection(capsys):
 
     hints = advise(d)
     assert any("256 CUs" in h[2] for h in hints)
+
+
+def test_gpu_concurrency_factor():
+    """Achieved kernel concurrency: duration-sum over busy-interval union."""
+    df = new_trace_df(4)
+    df["copyKind"] = 0
+    df["deviceId"] = 0
+    df["name"] = "k"
+    # two overlapping pairs: [0,1]+[0.5,1.5] and [3,4]+[3,4]
+    df["timestamp"] = [0.0, 0.5, 3.0, 3.0]
+    df["duration"] = [1.0, 1.0, 1.0, 1.0]
+    feats = []
+    profiles.gpu_profile(df, None, feats)
+    d = dict(feats)
+    # union = 1.5 + 1.0 = 2.5; total = 4.0 -> 1.6x
+    assert abs(d["gpu0_concurrency_factor"] - 1.6) < 1e-9
+
+
+def test_gpu_concurrency_serialized():
+    df = new_trace_df(2)
+    df["copyKind"] = 0
+    df["deviceId"] = 1
+    df["name"] = "k"
+    df["timestamp"] = [0.0, 2.0]
+    df["duration"] = [1.0, 1.0]
+    feats = []
+    profiles.gpu_profile(df, None, feats)
+    assert abs(dict(feats)["gpu1_concurrency_factor"] - 1.0) < 1e-9
