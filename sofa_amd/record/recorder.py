@@ -169,6 +169,17 @@ def build_target_env(cfg: SofaConfig) -> dict:
         mode = getattr(cfg, "gpu_tracer", "sdk")
         if mode == "lite" and not os.path.exists(lite):
             mode = "sdk"  # graceful fallback
+        if mode == "lite" and os.environ.get("ROCP_TOOL_LIBRARIES"):
+            # ROCr skips HSA_TOOLS_LIB tools when rocprofiler is registered
+            # (measured; see hsalite.cc) — a pre-set ROCP_TOOL_LIBRARIES
+            # (e.g. running under rocprofv3) would silently disable the lite
+            # collector, so fall back to the sdk path and say so
+            p.print_warning(
+                "ROCP_TOOL_LIBRARIES is set in the environment; the lite "
+                "collector cannot coexist with rocprofiler — using "
+                "--gpu_tracer sdk for this run"
+            )
+            mode = "sdk"
         if mode == "lite" and os.path.exists(lite):
             # HSA-level dispatch/copy tracer (lowest overhead, 2.9% measured
             # vs the SDK's 12.8% on ResNet-50 bs=64; hsalite/hsalite.cc).

@@ -89,3 +89,21 @@ def test_sofa_edr_triggers(tmp_path, monkeypatch):
     assert len(calls) == 1
     assert "sleep 5" in " ".join(calls[0])
     assert any(str(tmp_path / "edr") + "-forward" in c for c in calls[0])
+
+
+def test_lite_falls_back_when_rocprofiler_registered(monkeypatch):
+    """A pre-set ROCP_TOOL_LIBRARIES (e.g. running under rocprofv3) would
+    silently disable the lite collector — the recorder must fall back to the
+    sdk path with a warning."""
+    import os
+
+    from sofa_amd.config import SofaConfig
+    from sofa_amd.record.recorder import build_target_env
+
+    monkeypatch.setenv("ROCP_TOOL_LIBRARIES", "/opt/rocm/lib/librocprofv3-whatever.so")
+    cfg = SofaConfig(logdir="/tmp/x", gpu_tracer="lite")
+    env = build_target_env(cfg)
+    lite = "libsofahsalite.so"
+    assert lite not in env.get("HSA_TOOLS_LIB", "")
+    # sdk fallback appends the SDK collector after the user's tool
+    assert "libsofatracer.so" in env.get("ROCP_TOOL_LIBRARIES", "")
