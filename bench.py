@@ -43,7 +43,10 @@ def setup_tracer_env(logdir: str, tracer_mode: str) -> None:
     os.environ["SOFA_DEFER_START"] = "1"
     if tracer_mode == "lite":
         # HSA-level dispatch/copy tracer, SDK-free (rocprofiler's presence
-        # blocks hsalite's OnLoad — measured); RCCL args via debug log
+        # blocks hsalite's OnLoad — measured); RCCL args via debug log.
+        # A stray ROCP_TOOL_LIBRARIES would silently disable the collector:
+        # the bench owns its process env, so drop it.
+        os.environ.pop("ROCP_TOOL_LIBRARIES", None)
         prev_hsa = os.environ.get("HSA_TOOLS_LIB", "")
         if LITE not in prev_hsa:
             os.environ["HSA_TOOLS_LIB"] = LITE + ((" " + prev_hsa) if prev_hsa else "")
