@@ -126,6 +126,7 @@ def sofa_analyze(cfg: SofaConfig, pre: Optional[dict] = None) -> Dict[str, float
     comm_mod.comm_profile(logdir, df_gpu, features)
     comm_mod.rccl_link_attribution(logdir, df_rccl, topo, features, df_gpu=df_gpu)
     profiles.xgmi_measured_profile(logdir, features)
+    profiles.pc_hotspot_profile(logdir, features)
 
     # --- launch-latency / launch-bound analysis (corr-id join) ---
     try:

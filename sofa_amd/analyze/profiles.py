@@ -226,6 +226,41 @@ def diskstat_profile(logdir: str, features: Features) -> None:
     features.append(("disk_write_Bps", float(d["write_Bps"].mean())))
 
 
+def pc_hotspot_profile(logdir: str, features: Features) -> None:
+    """GPU instruction-level hotspots from PC samples (pcsamples.csv,
+    --pc_sampling): per-kernel sample share, SIMD-lane activity (divergence
+    signal, /64), and the hottest code-object offsets.  Beyond-reference
+    capability."""
+    path = os.path.join(logdir, "pcsamples.csv")
+    if not os.path.isfile(path):
+        return
+    try:
+        d = pd.read_csv(path)
+    except (OSError, pd.errors.ParserError):
+        return
+    if len(d) == 0:
+        return
+    total = len(d)
+    features.append(("pcsamples_total", float(total)))
+    features.append(("pcsamples_mean_active_lanes", float(d["active_lanes"].mean())))
+    print("\nGPU PC-sample hotspots (%d samples):" % total)
+    print("%-58s %8s %7s %9s" % ("kernel", "samples", "share", "lanes/64"))
+    g = d.groupby("kernel").agg(
+        samples=("offset", "size"), lanes=("active_lanes", "mean")
+    ).sort_values("samples", ascending=False)
+    for name, row in g.head(8).iterrows():
+        short = str(name) if len(str(name)) < 58 else str(name)[:55] + "..."
+        print("%-58s %8d %6.1f%% %9.1f" % (
+            short, row["samples"], 100.0 * row["samples"] / total, row["lanes"]))
+    top_kernel = g.index[0]
+    hot = d[d["kernel"] == top_kernel]
+    offs = hot.groupby("offset").size().sort_values(ascending=False).head(3)
+    print("  hottest offsets in %s:" % (str(top_kernel)[:60]))
+    for off, cnt in offs.items():
+        print("    +0x%x  %d samples (%.1f%% of kernel)" % (
+            int(off), cnt, 100.0 * cnt / len(hot)))
+
+
 def xgmi_measured_profile(logdir: str, features: Features) -> None:
     """Measured per-xGMI-link bandwidth from gpu_metrics HW accumulators
     (xgmi_counters.csv) — the ground truth the analytic ring model in

@@ -95,6 +95,9 @@ def build_parser() -> argparse.ArgumentParser:
                     help="per-request block-IO latency via tracefs "
                     "block_rq_issue/complete (blktrace parity)")
     ap.add_argument("--blkdev", default="", help="block device hint (implies --enable_blkio)")
+    ap.add_argument("--pc_sampling", action="store_true",
+                    help="GPU program-counter sampling (instruction-level "
+                    "hotspots; implies --gpu_tracer sdk; experimental SDK API)")
     ap.add_argument("--enable_kfd_trace", action="store_true",
                     help="trace KFD page-migrate/fault events (SVM memory pressure)")
     # preprocess
@@ -163,7 +166,8 @@ def cfg_from_args(args) -> SofaConfig:
         enable_gpu=not args.no_gpu,
         enable_gpu_hip_api=not args.no_hip_api,
         hip_api_full=args.hip_api_full,
-        gpu_tracer=args.gpu_tracer,
+        gpu_tracer=("sdk" if args.pc_sampling else args.gpu_tracer),
+        pc_sampling=args.pc_sampling,
         enable_blkio=args.enable_blkio,
         blkdev=args.blkdev,
         docker_image=args.docker_image,

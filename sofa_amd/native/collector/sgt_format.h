@@ -38,6 +38,7 @@ enum RecType : uint16_t {
   REC_DROP = 10,        // DropRec (buffer drops, should be 0 for lossless)
   REC_MARKER = 11,      // NameRec-shaped: roctx range/instant (id -> message)
   REC_KFD = 12,         // KfdRec: page migrate/fault/queue events
+  REC_PCSAMPLE = 13,    // PcSampleRec: GPU program-counter sample
 };
 
 // KFD event classes (KfdRec.op_class)
@@ -169,5 +170,17 @@ struct KfdRec {
   int32_t src_device; // migrate: source agent's logical id
   int32_t error_code;
 };  // 56 bytes
+
+struct PcSampleRec {
+  RecHeader h;              // REC_PCSAMPLE
+  uint64_t timestamp;       // rocprofiler clock ns
+  uint64_t corr_id;         // joins the owning kernel dispatch
+  uint64_t code_object_id;
+  uint64_t offset;          // PC offset within the code object
+  uint64_t exec_mask;       // active SIMD lanes at sample time
+  uint64_t dispatch_id;
+  uint32_t wave_in_group;
+  uint32_t device;          // logical GPU index
+};  // 64 bytes
 
 }  // namespace sgt

@@ -29,6 +29,7 @@
 //   SOFA_TRACE_ALLOC   1/0 (default 0)
 //   SOFA_GPU_BUFFER_MB per-process SDK buffer MiB (default 64)
 
+#include <rocprofiler-sdk/pc_sampling.h>
 #include <rocprofiler-sdk/registration.h>
 #include <rocprofiler-sdk/rocprofiler.h>
 #include <rocprofiler-sdk/rccl.h>
@@ -56,6 +57,9 @@ namespace {
 rocprofiler_client_id_t* g_client_id = nullptr;
 rocprofiler_context_id_t g_ctx = {0};
 rocprofiler_buffer_id_t g_buffer = {};
+rocprofiler_buffer_id_t g_pc_buffer = {};
+std::vector<rocprofiler_agent_id_t> g_gpu_agents;  // fill at write_agents
+bool g_pc_active = false;
 
 FILE* g_out = nullptr;
 std::mutex g_mutex;
@@ -426,6 +430,105 @@ void buffer_callback(rocprofiler_context_id_t, rocprofiler_buffer_id_t,
   }
 }
 
+// GPU program-counter samples (SOFA_PC_SAMPLING=1; experimental SDK API,
+// host-trap method): each sample -> PcSampleRec keyed by the dispatch's
+// correlation id, so preprocess attributes hotspots per kernel.  A feature
+// the reference never had (nvprof exposed no PC sampling to SOFA).
+void pc_buffer_callback(rocprofiler_context_id_t, rocprofiler_buffer_id_t,
+                        rocprofiler_record_header_t** headers,
+                        size_t num_headers, void*, uint64_t drop_count) {
+  if (drop_count > 0) {
+    sgt::DropRec d{};
+    d.h = {sgt::REC_DROP, sizeof(sgt::DropRec), 0};
+    d.dropped = drop_count;
+    write_raw(&d, sizeof(d));
+  }
+  std::vector<char> chunk;
+  chunk.reserve(num_headers * sizeof(sgt::PcSampleRec));
+  size_t n = 0;
+  for (size_t i = 0; i < num_headers; ++i) {
+    auto* header = headers[i];
+    if (header->category != ROCPROFILER_BUFFER_CATEGORY_PC_SAMPLING) continue;
+    if (header->kind != ROCPROFILER_PC_SAMPLING_RECORD_HOST_TRAP_V0_SAMPLE)
+      continue;
+    auto* r = static_cast<rocprofiler_pc_sampling_record_host_trap_v0_t*>(
+        header->payload);
+    sgt::PcSampleRec rec{};
+    rec.h = {sgt::REC_PCSAMPLE, sizeof(sgt::PcSampleRec), 0};
+    rec.timestamp = r->timestamp;
+    rec.corr_id = r->correlation_id.internal;
+    rec.code_object_id = r->pc.code_object_id;
+    rec.offset = r->pc.code_object_offset;
+    rec.exec_mask = r->exec_mask;
+    rec.dispatch_id = r->dispatch_id;
+    rec.wave_in_group = r->wave_in_group;
+    rec.device = 0;
+    chunk.insert(chunk.end(), (const char*) &rec,
+                 (const char*) &rec + sizeof(rec));
+    ++n;
+  }
+  if (!chunk.empty()) {
+    write_raw(chunk.data(), chunk.size());
+    g_n_records.fetch_add(n, std::memory_order_relaxed);
+  }
+}
+
+struct PcCfgPick {
+  bool found = false;
+  rocprofiler_pc_sampling_method_t method{};
+  rocprofiler_pc_sampling_unit_t unit{};
+  uint64_t interval = 0;
+};
+
+void configure_pc_sampling() {
+  size_t buffer_bytes = 8 << 20;
+  if (rocprofiler_create_buffer(g_ctx, buffer_bytes, buffer_bytes / 2,
+                                ROCPROFILER_BUFFER_POLICY_LOSSLESS,
+                                pc_buffer_callback, nullptr,
+                                &g_pc_buffer) != ROCPROFILER_STATUS_SUCCESS)
+    return;
+  uint64_t req_us = 200;  // default ~5 kHz host-trap
+  if (const char* v = getenv("SOFA_PC_SAMPLING_INTERVAL_US"); v && *v)
+    req_us = strtoull(v, nullptr, 10);
+  for (auto agent : g_gpu_agents) {
+    PcCfgPick pick;
+    rocprofiler_query_pc_sampling_agent_configurations(
+        agent,
+        [](const rocprofiler_pc_sampling_configuration_t* cfgs, size_t n,
+           void* ud) -> rocprofiler_status_t {
+          auto* p = static_cast<PcCfgPick*>(ud);
+          for (size_t i = 0; i < n; ++i) {
+            if (cfgs[i].method == ROCPROFILER_PC_SAMPLING_METHOD_HOST_TRAP) {
+              p->found = true;
+              p->method = cfgs[i].method;
+              p->unit = cfgs[i].unit;
+              p->interval = cfgs[i].min_interval;  // caller raises to request
+              return ROCPROFILER_STATUS_SUCCESS;
+            }
+          }
+          return ROCPROFILER_STATUS_SUCCESS;
+        },
+        &pick);
+    if (!pick.found) continue;
+    uint64_t interval = req_us * 1000;  // ns when unit == TIME
+    if (pick.unit != ROCPROFILER_PC_SAMPLING_UNIT_TIME)
+      interval = req_us;  // fall back to the raw request for cycle units
+    if (interval < pick.interval) interval = pick.interval;
+    auto st = rocprofiler_configure_pc_sampling_service(
+        g_ctx, agent, pick.method, pick.unit, interval, g_pc_buffer, 0);
+    if (st == ROCPROFILER_STATUS_SUCCESS) {
+      g_pc_active = true;
+      fprintf(stderr,
+              "[sofatracer] pc sampling active (agent %lx, interval %lu)\n",
+              (unsigned long) agent.handle, (unsigned long) interval);
+    } else {
+      fprintf(stderr, "[sofatracer] pc sampling unavailable (agent %lx): %s\n",
+              (unsigned long) agent.handle,
+              rocprofiler_get_status_string(st));
+    }
+  }
+}
+
 void write_agents() {
   rocprofiler_query_available_agents(
       ROCPROFILER_AGENT_INFO_VERSION_0,
@@ -438,6 +541,7 @@ void write_agents() {
                             ? a->logical_node_type_id
                             : -1;
           agent_device_map()[a->id.handle] = dev;
+          if (dev >= 0) g_gpu_agents.push_back(a->id);
           sgt::AgentRec rec{};
           rec.h = {sgt::REC_AGENT, sizeof(sgt::AgentRec), 0};
           rec.agent_handle = a->id.handle;
@@ -623,6 +727,7 @@ int tool_init(rocprofiler_client_finalize_t, void*) {
   rocprofiler_configure_callback_tracing_service(
       g_ctx, ROCPROFILER_CALLBACK_TRACING_MARKER_CORE_API, nullptr, 0,
       marker_callback, nullptr);
+  if (env_flag("SOFA_PC_SAMPLING", false)) configure_pc_sampling();
 
   auto cb_thread = rocprofiler_callback_thread_t{};
   if (rocprofiler_create_callback_thread(&cb_thread) ==
@@ -670,6 +775,8 @@ extern "C" unsigned long long sofa_tracer_event_count() {
 }
 
 extern "C" int sofa_tracer_active() { return g_out != nullptr; }
+
+extern "C" int sofa_tracer_pc_sampling_active() { return g_pc_active ? 1 : 0; }
 
 extern "C" rocprofiler_tool_configure_result_t*
 rocprofiler_configure(uint32_t version, const char* runtime_version,

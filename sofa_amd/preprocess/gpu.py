@@ -376,3 +376,57 @@ def sgt_to_markers(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.DataFrame
     df.sort_values("timestamp", inplace=True, kind="stable")
     df.reset_index(drop=True, inplace=True)
     return df
+
+
+def sgt_to_pcsamples(files: List[SgtFile], tb: Optional[TimeBase]) -> pd.DataFrame:
+    """GPU PC samples -> flat dataframe (pcsamples.csv), kernel-attributed.
+
+    Samples carry the kernel dispatch correlation id; joining against the
+    kernel records gives per-kernel instruction-level hotspots (offset is
+    the PC offset within the loaded code object).  active_lanes is the
+    popcount of the 64-bit exec mask — a direct divergence/occupancy
+    signal per sample.  Beyond-reference capability (nvprof exposed no PC
+    sampling to the reference)."""
+    frames = []
+    for sgt in files:
+        s = sgt.pcsamples
+        if not len(s):
+            continue
+        k = sgt.kernels
+        name_by_corr = {}
+        if len(k):
+            names = sgt.kernel_names
+            for corr, kid in zip(k["corr_id"], k["kernel_id"]):
+                name_by_corr[int(corr)] = names.get(int(kid), "kernel_%d" % kid)
+        corr = s["corr_id"].astype(np.int64)
+        uniq, inv = np.unique(corr, return_inverse=True)
+        uniq_names = np.array(
+            [demangle(name_by_corr.get(int(c), "unknown")) for c in uniq],
+            dtype=object,
+        )
+        # popcount of exec_mask, vectorized via a contiguous uint8 view
+        mask = np.ascontiguousarray(s["exec_mask"].astype(np.uint64))
+        lanes = (
+            np.unpackbits(mask.view(np.uint8).reshape(len(s), 8), axis=1)
+            .sum(axis=1)
+            .astype(np.int64)
+        )
+        frames.append(
+            pd.DataFrame(
+                {
+                    "timestamp": _timeline(tb, sgt, s["timestamp"]),
+                    "kernel": uniq_names[inv],
+                    "code_object_id": s["code_object_id"].astype(np.int64),
+                    "offset": s["offset"].astype(np.int64),
+                    "active_lanes": lanes,
+                    "dispatch_id": s["dispatch_id"].astype(np.int64),
+                    "wave_in_group": s["wave_in_group"].astype(np.int64),
+                    "pid": sgt.pid,
+                }
+            )
+        )
+    if not frames:
+        return pd.DataFrame()
+    out = pd.concat(frames, ignore_index=True)
+    out.sort_values("timestamp", inplace=True, kind="stable")
+    return out.reset_index(drop=True)

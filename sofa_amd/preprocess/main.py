@@ -210,6 +210,17 @@ def sofa_preprocess(cfg: SofaConfig) -> dict:
         write_trace_csv(df_marks, os.path.join(logdir, "markers.csv"))
         traces.append(SOFATrace(name="roctx_traces", title="roctx markers", color="Black", data=df_marks))
     result["df_markers"] = df_marks
+    # GPU PC samples (opt-in --pc_sampling; sdk collector)
+    try:
+        df_pc = gpu_mod.sgt_to_pcsamples(sgt_files, tb) if sgt_files else None
+        if df_pc is not None and len(df_pc):
+            df_pc.to_csv(os.path.join(logdir, "pcsamples.csv"), index=False)
+            p.print_info(f"{len(df_pc)} GPU PC samples -> pcsamples.csv")
+        result["df_pcsamples"] = df_pc
+    except Exception as e:
+        p.print_warning(f"pc sample parse failed: {e}")
+        result["df_pcsamples"] = None
+
     df_kfd = gpu_mod.sgt_to_kfdtrace(sgt_files, tb) if sgt_files else new_trace_df(0)
     if len(df_kfd):
         write_trace_csv(df_kfd, os.path.join(logdir, "kfdtrace.csv"))

@@ -31,6 +31,7 @@ REC_ALLOC = 9
 REC_DROP = 10
 REC_MARKER = 11
 REC_KFD = 12
+REC_PCSAMPLE = 13
 
 _HDR = [("type", "<u2"), ("size", "<u2"), ("_pad", "<u4")]
 
@@ -130,8 +131,23 @@ ALLOC_DTYPE = np.dtype(
     ]
 )
 
+PCSAMPLE_DTYPE = np.dtype(
+    _HDR
+    + [
+        ("timestamp", "<u8"),
+        ("corr_id", "<u8"),
+        ("code_object_id", "<u8"),
+        ("offset", "<u8"),
+        ("exec_mask", "<u8"),
+        ("dispatch_id", "<u8"),
+        ("wave_in_group", "<u4"),
+        ("device", "<u4"),
+    ]
+)
+
 FIXED_DTYPES = {
     REC_KERNEL: KERNEL_DTYPE,
+    REC_PCSAMPLE: PCSAMPLE_DTYPE,
     REC_KFD: KFD_DTYPE,
     REC_COPY: COPY_DTYPE,
     REC_HIPAPI: API_DTYPE,
@@ -152,6 +168,7 @@ class SgtFile:
     rccl: np.ndarray = field(default_factory=lambda: np.empty(0, RCCL_DTYPE))
     allocs: np.ndarray = field(default_factory=lambda: np.empty(0, ALLOC_DTYPE))
     kfd: np.ndarray = field(default_factory=lambda: np.empty(0, KFD_DTYPE))
+    pcsamples: np.ndarray = field(default_factory=lambda: np.empty(0, PCSAMPLE_DTYPE))
     kernel_names: Dict[int, str] = field(default_factory=dict)
     markers: List[tuple] = field(default_factory=list)  # (rocp_ns, message)
     opnames: Dict[tuple, str] = field(default_factory=dict)  # (kind, op) -> name
@@ -274,4 +291,5 @@ def parse_sgt(path: str) -> SgtFile:
     out.rccl = cat(REC_RCCL, RCCL_DTYPE)
     out.allocs = cat(REC_ALLOC, ALLOC_DTYPE)
     out.kfd = cat(REC_KFD, KFD_DTYPE)
+    out.pcsamples = cat(REC_PCSAMPLE, PCSAMPLE_DTYPE)
     return out

@@ -207,6 +207,8 @@ def build_target_env(cfg: SofaConfig) -> dict:
             env["SOFA_GPU_BUFFER_MB"] = str(cfg.gpu_ring_buffer_mb)
             if cfg.enable_kfd_trace:
                 env["SOFA_TRACE_KFD"] = "1"
+            if getattr(cfg, "pc_sampling", False):
+                env["SOFA_PC_SAMPLING"] = "1"
         else:
             p.print_warning("libsofatracer.so not built; GPU tracing disabled")
         if cfg.rccl_shim:

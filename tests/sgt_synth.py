@@ -19,6 +19,7 @@ from sofa_amd.preprocess.sgt import (
     REC_KERNEL,
     REC_KERNEL_NAME,
     REC_OPNAME,
+    REC_PCSAMPLE,
     REC_RCCL,
 )
 
@@ -125,6 +126,13 @@ class SgtWriter:
         a["comm"] = comm
         a["stream"] = stream
         self.buf += a.tobytes()
+
+    def pcsample(self, ts, corr, code_object_id=1, offset=0x100,
+                 exec_mask=(1 << 64) - 1, dispatch_id=0, wave=0, device=0):
+        self.buf += struct.pack(
+            "<HHIQQQQQQII", REC_PCSAMPLE, 64, 0, ts, corr, code_object_id,
+            offset, exec_mask, dispatch_id, wave, device,
+        )
 
     def write(self, path):
         with open(path, "wb") as f:

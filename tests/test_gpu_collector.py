@@ -123,3 +123,39 @@ def test_sofa_stat_torch_resnet_like(tmp_path):
     assert len(kernels) > 50, "expected many kernels from 3 training steps"
     # MIOpen/rocBLAS kernels appear with demangled-ish names
     assert kernels["name"].str.len().max() > 10
+
+
+def test_pc_sampling_hotspots(tmp_path):
+    """GPU PC sampling (--pc_sampling, experimental SDK API): samples land
+    in pcsamples.csv attributed to kernels.  Skips when the driver/stack
+    rejects host-trap sampling (the collector prints its verdict)."""
+    logdir = str(tmp_path / "log")
+    code = (
+        "import torch\n"
+        "x = torch.randn(4096, 4096, device='cuda')\n"
+        "for _ in range(60):\n"
+        "    x = (x @ x)\n"
+        "    x = x / x.norm()\n"
+        "torch.cuda.synchronize(); print('okay')\n"
+    )
+    r = subprocess.run(
+        [sys.executable, SOFA, "stat", f"{sys.executable} -c \"{code}\"",
+         "--logdir", logdir, "--pc_sampling"],
+        capture_output=True, text=True, timeout=900,
+    )
+    assert "Complete!!" in r.stdout, (r.stdout[-3000:], r.stderr[-2000:])
+    if "pc sampling active" not in r.stderr and "pc sampling active" not in r.stdout:
+        import pytest as _pytest
+
+        _pytest.skip("host-trap PC sampling not available on this stack: "
+                     + (r.stderr.splitlines()[-1] if r.stderr else "?"))
+    csv_path = os.path.join(logdir, "pcsamples.csv")
+    assert os.path.isfile(csv_path), "pc sampling active but no pcsamples.csv"
+    import pandas as pd
+
+    df = pd.read_csv(csv_path)
+    assert len(df) > 10, "too few PC samples for a 60-matmul loop"
+    # the GEMM must dominate the sample mass
+    top = df.groupby("kernel").size().sort_values(ascending=False)
+    assert any("Cijk" in str(k) or "gemm" in str(k).lower() for k in top.index[:2]), top.index[:3]
+    assert df["active_lanes"].between(0, 64).all()

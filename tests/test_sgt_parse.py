@@ -113,3 +113,52 @@ def test_markers_to_timeline(tmp_path):
     assert len(df) == 2
     assert df["name"].tolist() == ["roctx:step_begin", "roctx:step_end"]
     assert abs(df["timestamp"].iloc[0] - 2000.0005) < 1e-6
+
+
+def test_pcsample_records_roundtrip(tmp_path):
+    """PC-sample records: synth write -> parse -> kernel-attributed frame."""
+    from sgt_synth import SgtWriter
+    from sofa_amd.preprocess.gpu import sgt_to_pcsamples
+    from sofa_amd.preprocess.sgt import parse_sgt
+
+    w = SgtWriter()
+    w.kernel_name(5, "_Z8hot_loopv")
+    w.kernel(w.rocp_ns, w.rocp_ns + 1000, 5, corr=77)
+    for i in range(6):
+        w.pcsample(w.rocp_ns + 100 * i, corr=77, offset=0x40 + 16 * (i % 2),
+                   exec_mask=(1 << 32) - 1)
+    path = os.path.join(str(tmp_path), "gputrace_1.sgt")
+    w.write(path)
+    s = parse_sgt(path)
+    assert len(s.pcsamples) == 6
+    df = sgt_to_pcsamples([s], None)
+    assert len(df) == 6
+    assert (df["kernel"] == "hot_loop()").all()
+    assert (df["active_lanes"] == 32).all()  # half-divergent wave
+    assert set(df["offset"]) == {0x40, 0x50}
+
+
+def test_pc_hotspot_profile(tmp_path, capsys):
+    import pandas as pd
+
+    from sofa_amd.analyze.profiles import pc_hotspot_profile
+
+    pd.DataFrame(
+        {
+            "timestamp": [0.1] * 10,
+            "kernel": ["hot_kernel"] * 8 + ["cold_kernel"] * 2,
+            "code_object_id": 1,
+            "offset": [0x40] * 6 + [0x80] * 2 + [0x10] * 2,
+            "active_lanes": [64] * 8 + [8] * 2,
+            "dispatch_id": 0,
+            "wave_in_group": 0,
+            "pid": 1,
+        }
+    ).to_csv(os.path.join(str(tmp_path), "pcsamples.csv"), index=False)
+    feats = []
+    pc_hotspot_profile(str(tmp_path), feats)
+    d = dict(feats)
+    assert d["pcsamples_total"] == 10
+    out = capsys.readouterr().out
+    assert "hot_kernel" in out and "80.0%" in out
+    assert "+0x40" in out
