@@ -509,7 +509,13 @@ void configure_pc_sampling() {
           return ROCPROFILER_STATUS_SUCCESS;
         },
         &pick);
-    if (!pick.found) continue;
+    if (!pick.found) {
+      fprintf(stderr,
+              "[sofatracer] pc sampling unavailable (agent %lx): no host-trap "
+              "configuration offered by the driver\n",
+              (unsigned long) agent.handle);
+      continue;
+    }
     uint64_t interval = req_us * 1000;  // ns when unit == TIME
     if (pick.unit != ROCPROFILER_PC_SAMPLING_UNIT_TIME)
       interval = req_us;  // fall back to the raw request for cycle units
