@@ -449,19 +449,32 @@ void pc_buffer_callback(rocprofiler_context_id_t, rocprofiler_buffer_id_t,
   for (size_t i = 0; i < num_headers; ++i) {
     auto* header = headers[i];
     if (header->category != ROCPROFILER_BUFFER_CATEGORY_PC_SAMPLING) continue;
-    if (header->kind != ROCPROFILER_PC_SAMPLING_RECORD_HOST_TRAP_V0_SAMPLE)
-      continue;
-    auto* r = static_cast<rocprofiler_pc_sampling_record_host_trap_v0_t*>(
-        header->payload);
     sgt::PcSampleRec rec{};
     rec.h = {sgt::REC_PCSAMPLE, sizeof(sgt::PcSampleRec), 0};
-    rec.timestamp = r->timestamp;
-    rec.corr_id = r->correlation_id.internal;
-    rec.code_object_id = r->pc.code_object_id;
-    rec.offset = r->pc.code_object_offset;
-    rec.exec_mask = r->exec_mask;
-    rec.dispatch_id = r->dispatch_id;
-    rec.wave_in_group = r->wave_in_group;
+    if (header->kind == ROCPROFILER_PC_SAMPLING_RECORD_HOST_TRAP_V0_SAMPLE) {
+      auto* r = static_cast<rocprofiler_pc_sampling_record_host_trap_v0_t*>(
+          header->payload);
+      rec.timestamp = r->timestamp;
+      rec.corr_id = r->correlation_id.internal;
+      rec.code_object_id = r->pc.code_object_id;
+      rec.offset = r->pc.code_object_offset;
+      rec.exec_mask = r->exec_mask;
+      rec.dispatch_id = r->dispatch_id;
+      rec.wave_in_group = r->wave_in_group;
+    } else if (header->kind ==
+               ROCPROFILER_PC_SAMPLING_RECORD_STOCHASTIC_V0_SAMPLE) {
+      auto* r = static_cast<rocprofiler_pc_sampling_record_stochastic_v0_t*>(
+          header->payload);
+      rec.timestamp = r->timestamp;
+      rec.corr_id = r->correlation_id.internal;
+      rec.code_object_id = r->pc.code_object_id;
+      rec.offset = r->pc.code_object_offset;
+      rec.exec_mask = r->exec_mask;
+      rec.dispatch_id = r->dispatch_id;
+      rec.wave_in_group = r->wave_in_group;
+    } else {
+      continue;
+    }
     rec.device = 0;
     chunk.insert(chunk.end(), (const char*) &rec,
                  (const char*) &rec + sizeof(rec));
@@ -498,13 +511,19 @@ void configure_pc_sampling() {
            void* ud) -> rocprofiler_status_t {
           auto* p = static_cast<PcCfgPick*>(ud);
           for (size_t i = 0; i < n; ++i) {
-            if (cfgs[i].method == ROCPROFILER_PC_SAMPLING_METHOD_HOST_TRAP) {
-              p->found = true;
-              p->method = cfgs[i].method;
-              p->unit = cfgs[i].unit;
-              p->interval = cfgs[i].min_interval;  // caller raises to request
-              return ROCPROFILER_STATUS_SUCCESS;
-            }
+            bool host_trap =
+                cfgs[i].method == ROCPROFILER_PC_SAMPLING_METHOD_HOST_TRAP;
+            bool stochastic =
+                cfgs[i].method == ROCPROFILER_PC_SAMPLING_METHOD_STOCHASTIC;
+            if (!host_trap && !stochastic) continue;
+            // prefer host-trap (time-based, simple); stochastic is the
+            // MI300+/gfx950 hardware sampler
+            if (p->found && p->method == ROCPROFILER_PC_SAMPLING_METHOD_HOST_TRAP)
+              continue;
+            p->found = true;
+            p->method = cfgs[i].method;
+            p->unit = cfgs[i].unit;
+            p->interval = cfgs[i].min_interval;  // caller raises to request
           }
           return ROCPROFILER_STATUS_SUCCESS;
         },
@@ -517,8 +536,11 @@ void configure_pc_sampling() {
       continue;
     }
     uint64_t interval = req_us * 1000;  // ns when unit == TIME
-    if (pick.unit != ROCPROFILER_PC_SAMPLING_UNIT_TIME)
-      interval = req_us;  // fall back to the raw request for cycle units
+    if (pick.unit != ROCPROFILER_PC_SAMPLING_UNIT_TIME) {
+      // cycles/instructions unit (stochastic): use a power of two around
+      // ~1M cycles (~0.5 ms at 2 GHz) unless the minimum is higher
+      interval = 1u << 20;
+    }
     if (interval < pick.interval) interval = pick.interval;
     auto st = rocprofiler_configure_pc_sampling_service(
         g_ctx, agent, pick.method, pick.unit, interval, g_pc_buffer, 0);
