@@ -530,8 +530,8 @@ void configure_pc_sampling() {
         &pick);
     if (!pick.found) {
       fprintf(stderr,
-              "[sofatracer] pc sampling unavailable (agent %lx): no host-trap "
-              "configuration offered by the driver\n",
+              "[sofatracer] pc sampling unavailable (agent %lx): driver offers no "
+              "host-trap or stochastic configuration\n",
               (unsigned long) agent.handle);
       continue;
     }
