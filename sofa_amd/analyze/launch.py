@@ -73,7 +73,9 @@ def launch_latency_profile(sgt_files, features: List[Tuple[str, float]]) -> None
     p50, p95 = np.percentile(lat, [50, 95]) / 1e3
     features.append(("launch_latency_us_p50", float(p50)))
     features.append(("launch_latency_us_p95", float(p95)))
-    print("\nLaunch latency (API return -> kernel start): p50 %.1f us, p95 %.1f us over %d matched launches"
+    print("\nEnqueue-to-start delay: p50 %.1f us, p95 %.1f us over %d matched "
+          "launches (includes enqueue-ahead backlog: large values in an "
+          "unsynchronized loop mean the CPU runs ahead, not launch overhead)"
           % (p50, p95, len(lat)))
     if len(gaps):
         idle_ratio = float(gaps.sum()) / max(
