@@ -163,3 +163,31 @@ def test_duration_timebox(tmp_path, native_built):
     took = _time.time() - t0
     assert "Complete!!" in r.stdout, r.stderr
     assert took < 25, f"duration not enforced ({took:.1f}s)"
+
+
+def test_diff_verb_end_to_end(tmp_path):
+    """`sofa diff` through the CLI: two recorded-with-swarms logdirs ->
+    swarm_diff.csv (reference bin/sofa:338-350 verb path)."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sofa = os.path.join(repo, "bin", "sofa")
+    logs = []
+    for i, count in enumerate((6, 6)):
+        logdir = str(tmp_path / f"log{i}")
+        r = subprocess.run(
+            [sys.executable, sofa, "stat",
+             f"dd if=/dev/zero of={tmp_path}/d{i}.out bs=8M count={count}",
+             "--logdir", logdir, "--enable_swarms", "--no_gpu"],
+            capture_output=True, text=True, timeout=300, cwd=repo,
+        )
+        assert "Complete!!" in r.stdout, (r.stdout[-2000:], r.stderr[-1500:])
+        logs.append(logdir)
+    r = subprocess.run(
+        [sys.executable, sofa, "diff", "--base_logdir", logs[0],
+         "--match_logdir", logs[1], "--logdir", logs[1], "--skip_preprocess"],
+        capture_output=True, text=True, timeout=300, cwd=repo,
+    )
+    assert r.returncode == 0, (r.stdout[-2000:], r.stderr[-1500:])
+    assert os.path.isfile(os.path.join(logs[1], "swarm_diff.csv")), r.stdout[-1500:]
