@@ -90,7 +90,12 @@ def sofa_swarm_diff(cfg: SofaConfig) -> pd.DataFrame:
     intersection_rate = matched / max(len(base), 1)
     print(out.to_string(index=False))
     print("caption intersection rate: %.2f" % intersection_rate)
-    dest = os.path.join(cfg.logdir if os.path.isdir(cfg.logdir) else ".", "swarm_diff.csv")
+    # reference wrote into its default sofalog/; falling back to cwd
+    # polluted the invoking directory — prefer the match logdir instead
+    dest_dir = cfg.logdir if os.path.isdir(cfg.logdir) else cfg.match_logdir
+    if not os.path.isdir(dest_dir):
+        dest_dir = "."
+    dest = os.path.join(dest_dir, "swarm_diff.csv")
     os.makedirs(os.path.dirname(dest) or ".", exist_ok=True)
     out.to_csv(dest, index=False)
     p.print_progress(f"wrote {dest}")
