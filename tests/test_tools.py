@@ -107,3 +107,17 @@ def test_lite_falls_back_when_rocprofiler_registered(monkeypatch):
     assert lite not in env.get("HSA_TOOLS_LIB", "")
     # sdk fallback appends the SDK collector after the user's tool
     assert "libsofatracer.so" in env.get("ROCP_TOOL_LIBRARIES", "")
+
+
+def test_top_once_cpu():
+    """`sofa top --once` renders one combined refresh without a GPU."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "bin", "sofa"), "top", "--once"],
+        capture_output=True, text=True, timeout=120, cwd=repo,
+    )
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert "CPU" in r.stdout
