@@ -263,6 +263,11 @@ constexpr uint32_t kSlotsPerQueue = 8192;
 // through an OpNameRec written at init)
 constexpr uint32_t kAqlSubmitOp = 60000;
 bool g_submit_spans = true;  // SOFA_LITE_SUBMIT_SPANS=0 to disable
+// SOFA_LITE_SAMPLE=N: attach a timing signal to every Nth dispatch only
+// (1 = every dispatch, the default).  Overhead scales ~1/N while the
+// submit spans still record every launch — the always-on production dial.
+uint32_t g_sample_every = 1;
+std::atomic<uint64_t> g_sample_counter{0};
 
 std::atomic<uint64_t> g_pool_exhausted{0};
 
@@ -366,6 +371,10 @@ void on_submit(const void* pkts, uint64_t pkt_count, uint64_t /*user_pkt_index*/
       g_stats.skipped_has_signal.fetch_add(1, std::memory_order_relaxed);
       continue;
     }
+    if (g_sample_every > 1 &&
+        (g_sample_counter.fetch_add(1, std::memory_order_relaxed) %
+         g_sample_every) != 0)
+      continue;  // sampled-timing mode: this dispatch goes untimed
 
     // acquire a slot + publish to the in-flight list in ONE critical
     // section (an unpublished-packet slot is harmless to the reaper: its
@@ -993,6 +1002,10 @@ bool OnLoad(void* table_v, uint64_t runtime_version, uint64_t failed_tool_count,
   }
   g_replace_signals = env_flag("SOFA_LITE_REPLACE_SIGNALS", false);
   g_submit_spans = env_flag("SOFA_LITE_SUBMIT_SPANS", true);
+  if (const char* v = getenv("SOFA_LITE_SAMPLE"); v && *v) {
+    unsigned long n = strtoul(v, nullptr, 10);
+    if (n >= 1) g_sample_every = (uint32_t) n;
+  }
   g_armed.store(!env_flag("SOFA_DEFER_START", false));
 
   open_output();

@@ -71,6 +71,8 @@ def configs(logdir):
         ("lite-full", {**base, "HSA_TOOLS_LIB": lite}),
         ("lite-rs", {**base, "HSA_TOOLS_LIB": lite, "SOFA_LITE_REPLACE_SIGNALS": "1"}),
         ("lite-nospans", {**base, "HSA_TOOLS_LIB": lite, "SOFA_LITE_SUBMIT_SPANS": "0"}),
+        ("lite-sample4", {**base, "HSA_TOOLS_LIB": lite, "SOFA_LITE_SAMPLE": "4"}),
+        ("lite-sample16", {**base, "HSA_TOOLS_LIB": lite, "SOFA_LITE_SAMPLE": "16"}),
         ("sdk-null", {**base, "ROCP_TOOL_LIBRARIES": sdk, "SOFA_NULL_SINK": "1",
                       "SOFA_TRACE_HIP_API": "1", "SOFA_TRACE_RCCL": "1"}),
         ("sdk-full", {**base, "ROCP_TOOL_LIBRARIES": sdk,
