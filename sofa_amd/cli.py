@@ -95,6 +95,10 @@ def build_parser() -> argparse.ArgumentParser:
                     help="per-request block-IO latency via tracefs "
                     "block_rq_issue/complete (blktrace parity)")
     ap.add_argument("--blkdev", default="", help="block device hint (implies --enable_blkio)")
+    ap.add_argument("--gpu_sample", type=int, default=1, metavar="N",
+                    help="lite collector: time every Nth dispatch "
+                    "(overhead ~3.4%%/N; submit spans keep full launch "
+                    "coverage; 1 = every dispatch)")
     ap.add_argument("--pc_sampling", action="store_true",
                     help="GPU program-counter sampling (instruction-level "
                     "hotspots; implies --gpu_tracer sdk; experimental SDK API)")
@@ -168,6 +172,7 @@ def cfg_from_args(args) -> SofaConfig:
         hip_api_full=args.hip_api_full,
         gpu_tracer=("sdk" if args.pc_sampling else args.gpu_tracer),
         pc_sampling=args.pc_sampling,
+        gpu_sample=args.gpu_sample,
         enable_blkio=args.enable_blkio,
         blkdev=args.blkdev,
         docker_image=args.docker_image,

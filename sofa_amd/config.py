@@ -98,6 +98,7 @@ class SofaConfig:
     rccl_shim: bool = False            # LD_PRELOAD interposer (fallback path)
     enable_kfd_trace: bool = False     # page-migrate/fault events
     pc_sampling: bool = False          # GPU PC sampling (sdk collector, experimental)
+    gpu_sample: int = 1                # lite: time every Nth dispatch (overhead ~1/N)
     gpu_ring_buffer_mb: int = 64       # collector buffer size per process
     blkdev: str = ""                   # block device for blktrace-like stats
     enable_blkio: bool = False         # tracefs block_rq_issue/complete per-IO tracing

@@ -190,6 +190,8 @@ def build_target_env(cfg: SofaConfig) -> dict:
             prev_hsa = env.get("HSA_TOOLS_LIB", "")
             env["HSA_TOOLS_LIB"] = lite + ((" " + prev_hsa) if prev_hsa else "")
             env["SOFA_LOGDIR"] = os.path.abspath(cfg.logdir)
+            if getattr(cfg, "gpu_sample", 1) > 1:
+                env["SOFA_LITE_SAMPLE"] = str(cfg.gpu_sample)
             if cfg.enable_rccl_trace and "NCCL_DEBUG" not in env:
                 env["NCCL_DEBUG"] = "INFO"
                 env["NCCL_DEBUG_SUBSYS"] = "COLL"
