@@ -121,3 +121,13 @@ def test_top_once_cpu():
     )
     assert r.returncode == 0, r.stderr[-1500:]
     assert "CPU" in r.stdout
+
+
+def test_gpu_sample_env_plumbing():
+    from sofa_amd.config import SofaConfig
+    from sofa_amd.record.recorder import build_target_env
+
+    cfg = SofaConfig(logdir="/tmp/x", gpu_tracer="lite", gpu_sample=16)
+    env = build_target_env(cfg)
+    if "libsofahsalite.so" in env.get("HSA_TOOLS_LIB", ""):
+        assert env.get("SOFA_LITE_SAMPLE") == "16"
