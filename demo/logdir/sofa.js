@@ -126,6 +126,17 @@ SofaPlot.prototype.draw = function () {
   ctx.fillStyle = "#444";
   ctx.fillText(this.opts.xlabel || "time (s)", W / 2 - 20, H - 6);
 
+  // cross-page time cursor (set by click on any page; shared via
+  // localStorage so every open sofaboard page shows the same instant)
+  var cur = parseFloat(localStorage.getItem("sofa_cursor_t"));
+  if (isFinite(cur) && cur >= r.x0 && cur <= r.x1) {
+    ctx.strokeStyle = "#d22"; ctx.lineWidth = 1.5;
+    ctx.beginPath(); ctx.moveTo(xs(cur), m.t); ctx.lineTo(xs(cur), H - m.b); ctx.stroke();
+    ctx.fillStyle = "#d22";
+    ctx.fillText("t=" + cur.toFixed(4), xs(cur) + 4, m.t + 12);
+    ctx.lineWidth = 1;
+  }
+
   // series
   this.series.forEach(function (s) {
     if (!s.visible) return;
@@ -156,7 +167,17 @@ SofaPlot.prototype._bindEvents = function () {
     if (dragStart === null) return;
     var a = dragStart, b = e.offsetX;
     dragStart = null;
-    if (Math.abs(a - b) < 8) { return; }
+    if (Math.abs(a - b) < 8) {
+      // plain click: place the shared time cursor at this instant
+      var rr = self._range;
+      var t = rr.x0 + (b - self.margin.l) /
+        (self.canvas.width - self.margin.l - self.margin.r) * (rr.x1 - rr.x0);
+      if (isFinite(t)) {
+        localStorage.setItem("sofa_cursor_t", String(t));
+        self.draw();
+      }
+      return;
+    }
     var r = self._range;
     var inv = function (px) {
       return r.x0 + (px - self.margin.l) / (self.canvas.width - self.margin.l - self.margin.r) * (r.x1 - r.x0);
@@ -164,7 +185,15 @@ SofaPlot.prototype._bindEvents = function () {
     self.zoom = { x0: inv(Math.min(a, b)), x1: inv(Math.max(a, b)) };
     self.draw();
   });
-  this.canvas.addEventListener("dblclick", function () { self.zoom = null; self.draw(); });
+  this.canvas.addEventListener("dblclick", function () {
+    self.zoom = null;
+    localStorage.removeItem("sofa_cursor_t");
+    self.draw();
+  });
+  // other pages moved the cursor -> redraw live
+  window.addEventListener("storage", function (ev) {
+    if (ev.key === "sofa_cursor_t") self.draw();
+  });
   this.canvas.addEventListener("mousemove", function (e) {
     var tip = document.getElementById(self.opts.tooltipId || "");
     if (!tip || !self._xs) return;
